@@ -1,0 +1,992 @@
+#!/usr/bin/env python3
+"""Golden-fixture generator for the BLS12-381 hot path.
+
+A self-contained big-int BLS12-381 implementation (min_pk scheme, RFC 9380
+hash-to-curve BLS12381G2_XMD:SHA-256_SSWU_RO_) used ONLY to:
+  1. validate itself against the reference's in-repo interop vectors
+     (/root/reference/common/eth2_interop_keypairs/tests/generation.rs:10-45 —
+     the 10 decimal privkeys and 10 base64 compressed pubkeys are embedded
+     below as test DATA, with that citation);
+  2. emit tests/golden/bls_fixtures.json — golden vectors that pin the C
+     oracle (and the HIP path) offline;
+  3. emit oracle/bls_consts.h — numeric constants (Montgomery parameters,
+     generators, SSWU/isogeny/psi/Frobenius constants) that BOTH the C oracle
+     and the HIP kernels consume, each validated by numeric asserts here.
+
+Semantics restated from /root/reference/crypto/bls/src/impls/blst.rs:15-16
+(DST, RAND_BITS), :37-119 (batch equation + rejection rules),
+generic_public_key.rs:12-21 / generic_signature.rs:15-26 (encodings),
+eth2_interop_keypairs/src/lib.rs:40-55 (keygen), and RFC 9380 (public spec).
+
+Run: python3 tests/golden/gen_bls_fixtures.py   (regenerates both outputs)
+"""
+import base64
+import hashlib
+import json
+import os
+import sys
+
+# ---------------------------------------------------------------- constants
+P = 0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFAAAB
+R = 0x73EDA753299D7D483339D80809A1D80553BDA402FFFE5BFEFFFFFFFF00000001
+X_PARAM = -0xD201000000010000  # BLS parameter (negative)
+H_EFF_ABS = 0xD201000000010000
+
+assert P % 4 == 3 and P % 6 == 1
+assert P.bit_length() == 381 and R.bit_length() == 255
+
+G1X = 0x17F1D3A73197D7942695638C4FA9AC0FC3688C4F9774B905A14E3A3F171BAC586C55E83FF97A1AEFFB3AF00ADB22C6BB
+G1Y = 0x08B3F481E3AAA0F1A09E30ED741D8AE4FCF5E095D5D00AF600DB18CB2C04B3EDD03CC744A2888AE40CAA232946C5E7E1
+G2X = (
+    0x024AA2B2F08F0A91260805272DC51051C6E47AD4FA403B02B4510B647AE3D1770BAC0326A805BBEFD48056C8C121BDB8,
+    0x13E02B6052719F607DACD3A088274F65596BD0D09920B61AB5DA61BBDC7F5049334CF11213945D57E5AC7D055D042B7E,
+)
+G2Y = (
+    0x0CE5D527727D6E118CC9CDC6DA2E351AADFD9BAA8CBDD3A76D429A695160D12C923AC9CC3BACA289E193548608B82801,
+    0x0606C4A02EA734CC32ACD2B02BC28B99CB3E287E85A763AF267492AB572E99AB3F370D275CEC1DA1AAA9075FF05F79BE,
+)
+
+DST = b"BLS_SIG_BLS12381G2_XMD:SHA-256_SSWU_RO_POP_"  # blst.rs:15
+
+# ---------------------------------------------------------------- Fp / Fp2
+
+
+def finv(a):
+    return pow(a, P - 2, P)
+
+
+def fp_sqrt(a):
+    s = pow(a, (P + 1) // 4, P)
+    return s if s * s % P == a % P else None
+
+
+# Fp2 = Fp[u]/(u^2+1), elements (c0, c1)
+def f2add(a, b):
+    return ((a[0] + b[0]) % P, (a[1] + b[1]) % P)
+
+
+def f2sub(a, b):
+    return ((a[0] - b[0]) % P, (a[1] - b[1]) % P)
+
+
+def f2neg(a):
+    return ((-a[0]) % P, (-a[1]) % P)
+
+
+def f2mul(a, b):
+    return (
+        (a[0] * b[0] - a[1] * b[1]) % P,
+        (a[0] * b[1] + a[1] * b[0]) % P,
+    )
+
+
+def f2sq(a):
+    return f2mul(a, a)
+
+
+def f2smul(a, k):
+    return (a[0] * k % P, a[1] * k % P)
+
+
+def f2inv(a):
+    n = finv((a[0] * a[0] + a[1] * a[1]) % P)
+    return (a[0] * n % P, (-a[1] * n) % P)
+
+
+def f2conj(a):
+    return (a[0], (-a[1]) % P)
+
+
+def f2pow(a, e):
+    r = (1, 0)
+    while e:
+        if e & 1:
+            r = f2mul(r, a)
+        a = f2sq(a)
+        e >>= 1
+    return r
+
+
+F2_ZERO, F2_ONE = (0, 0), (1, 0)
+XI = (1, 1)  # the sextic non-residue u+1 (tower constant)
+
+
+def f2sqrt(a):
+    """sqrt in Fp2 for p=3 mod 4 via the norm method; None if non-square."""
+    if a == F2_ZERO:
+        return F2_ZERO
+    a0, a1 = a
+    if a1 == 0:
+        s = fp_sqrt(a0)
+        if s is not None:
+            return (s, 0)
+        s = fp_sqrt((-a0) % P)
+        assert s is not None
+        return (0, s)
+    n = (a0 * a0 + a1 * a1) % P
+    s = fp_sqrt(n)
+    if s is None:
+        return None
+    inv2 = finv(2)
+    d = (a0 + s) * inv2 % P
+    x0 = fp_sqrt(d)
+    if x0 is None:
+        d = (a0 - s) * inv2 % P
+        x0 = fp_sqrt(d)
+        if x0 is None:
+            return None
+    x1 = a1 * finv(2 * x0 % P) % P
+    cand = (x0, x1)
+    return cand if f2sq(cand) == (a0 % P, a1 % P) else None
+
+
+# ---------------------------------------------------------------- curves
+# G1: y^2 = x^3 + 4 over Fp ; G2: y^2 = x^3 + 4(u+1) over Fp2.
+B1 = 4
+B2 = f2smul(XI, 4)
+
+INF = None  # affine infinity
+
+
+def g1_add(p, q):
+    if p is None:
+        return q
+    if q is None:
+        return p
+    x1, y1 = p
+    x2, y2 = q
+    if x1 == x2:
+        if (y1 + y2) % P == 0:
+            return None
+        lam = 3 * x1 * x1 * finv(2 * y1 % P) % P
+    else:
+        lam = (y2 - y1) * finv((x2 - x1) % P) % P
+    x3 = (lam * lam - x1 - x2) % P
+    y3 = (lam * (x1 - x3) - y1) % P
+    return (x3, y3)
+
+
+def g1_mul(k, p):
+    r = None
+    k %= 2**512  # scalars are plain integers
+    while k:
+        if k & 1:
+            r = g1_add(r, p)
+        p = g1_add(p, p)
+        k >>= 1
+    return r
+
+
+def g1_neg(p):
+    return None if p is None else (p[0], (-p[1]) % P)
+
+
+def g2_add(p, q):
+    if p is None:
+        return q
+    if q is None:
+        return p
+    x1, y1 = p
+    x2, y2 = q
+    if x1 == x2:
+        if f2add(y1, y2) == F2_ZERO:
+            return None
+        lam = f2mul(f2smul(f2sq(x1), 3), f2inv(f2smul(y1, 2)))
+    else:
+        lam = f2mul(f2sub(y2, y1), f2inv(f2sub(x2, x1)))
+    x3 = f2sub(f2sub(f2sq(lam), x1), x2)
+    y3 = f2sub(f2mul(lam, f2sub(x1, x3)), y1)
+    return (x3, y3)
+
+
+def g2_mul(k, p):
+    r = None
+    while k:
+        if k & 1:
+            r = g2_add(r, p)
+        p = g2_add(p, p)
+        k >>= 1
+    return r
+
+
+def g2_neg(p):
+    return None if p is None else (p[0], f2neg(p[1]))
+
+
+def on_g1(p):
+    return p is None or (p[1] * p[1] - p[0] ** 3 - B1) % P == 0
+
+
+def on_g2(p):
+    return p is None or f2sub(f2sq(p[1]), f2add(f2mul(f2sq(p[0]), p[0]), B2)) == F2_ZERO
+
+
+assert on_g1((G1X, G1Y)), "G1 generator not on curve — constant typo"
+assert on_g2((G2X, G2Y)), "G2 generator not on curve — constant typo"
+assert g1_mul(R, (G1X, G1Y)) is None, "G1 generator order"
+assert g2_mul(R, (G2X, G2Y)) is None, "G2 generator order"
+
+# ------------------------------------------------------- serialization (ZCash)
+HALF_P = (P - 1) // 2
+
+
+def g1_compress(p):
+    if p is None:
+        return bytes([0xC0] + [0] * 47)
+    x, y = p
+    b = bytearray(x.to_bytes(48, "big"))
+    b[0] |= 0x80
+    if y > HALF_P:
+        b[0] |= 0x20
+    return bytes(b)
+
+
+def g1_uncompressed(p):
+    if p is None:
+        b = bytearray(96)
+        b[0] = 0x40
+        return bytes(b)
+    return p[0].to_bytes(48, "big") + p[1].to_bytes(48, "big")
+
+
+def g1_decompress(b):
+    """Returns point, or raises ValueError (malformed)."""
+    if len(b) != 48:
+        raise ValueError("len")
+    flags = b[0]
+    if not flags & 0x80:
+        raise ValueError("not compressed")
+    if flags & 0x40:
+        if flags & 0x20 or any(b[1:]) or (b[0] & 0x3F):
+            raise ValueError("bad infinity")
+        return None
+    x = int.from_bytes(bytes([b[0] & 0x1F]) + b[1:], "big")
+    if x >= P:
+        raise ValueError("x >= p")
+    y = fp_sqrt((x**3 + B1) % P)
+    if y is None:
+        raise ValueError("not on curve")
+    if (y > HALF_P) != bool(flags & 0x20):
+        y = P - y
+    return (x, y)
+
+
+def g2_compress(p):
+    if p is None:
+        return bytes([0xC0] + [0] * 95)
+    (x0, x1), (y0, y1) = p
+    b = bytearray(x1.to_bytes(48, "big") + x0.to_bytes(48, "big"))
+    b[0] |= 0x80
+    if (y1 != 0 and y1 > HALF_P) or (y1 == 0 and y0 > HALF_P):
+        b[0] |= 0x20
+    return bytes(b)
+
+
+def g2_uncompressed(p):
+    if p is None:
+        b = bytearray(192)
+        b[0] = 0x40
+        return bytes(b)
+    (x0, x1), (y0, y1) = p
+    return (
+        x1.to_bytes(48, "big")
+        + x0.to_bytes(48, "big")
+        + y1.to_bytes(48, "big")
+        + y0.to_bytes(48, "big")
+    )
+
+
+def g2_decompress(b):
+    if len(b) != 96:
+        raise ValueError("len")
+    flags = b[0]
+    if not flags & 0x80:
+        raise ValueError("not compressed")
+    if flags & 0x40:
+        if flags & 0x20 or any(b[1:]) or (b[0] & 0x3F):
+            raise ValueError("bad infinity")
+        return None
+    x1 = int.from_bytes(bytes([b[0] & 0x1F]) + b[1:48], "big")
+    x0 = int.from_bytes(b[48:96], "big")
+    if x0 >= P or x1 >= P:
+        raise ValueError("x >= p")
+    x = (x0, x1)
+    y = f2sqrt(f2add(f2mul(f2sq(x), x), B2))
+    if y is None:
+        raise ValueError("not on curve")
+    y0, y1 = y
+    big = (y1 != 0 and y1 > HALF_P) or (y1 == 0 and y0 > HALF_P)
+    if big != bool(flags & 0x20):
+        y = f2neg(y)
+    return (x, y)
+
+
+# ------------------------------------------------------- hash-to-curve (RFC 9380)
+def expand_message_xmd(msg, dst, length):
+    ell = (length + 31) // 32
+    assert ell <= 255 and len(dst) <= 255
+    dst_prime = dst + bytes([len(dst)])
+    z_pad = b"\x00" * 64
+    l_i_b = length.to_bytes(2, "big")
+    b0 = hashlib.sha256(z_pad + msg + l_i_b + b"\x00" + dst_prime).digest()
+    bvals = [hashlib.sha256(b0 + b"\x01" + dst_prime).digest()]
+    for i in range(2, ell + 1):
+        t = bytes(a ^ b for a, b in zip(b0, bvals[-1]))
+        bvals.append(hashlib.sha256(t + bytes([i]) + dst_prime).digest())
+    return b"".join(bvals)[:length]
+
+
+def hash_to_field_fp2(msg, count):
+    L = 64
+    u = expand_message_xmd(msg, DST, count * 2 * L)
+    out = []
+    for i in range(count):
+        e = []
+        for j in range(2):
+            off = L * (j + i * 2)
+            e.append(int.from_bytes(u[off : off + L], "big") % P)
+        out.append((e[0], e[1]))
+    return out
+
+
+# SSWU on E2': y^2 = x^3 + A'x + B' with A'=240u, B'=1012(1+u), Z=-(2+u)
+A_P = (0, 240)
+B_P = (1012, 1012)
+Z_SSWU = ((-2) % P, (-1) % P)
+
+
+def sgn0_fp2(x):
+    s0 = x[0] % 2
+    z0 = x[0] == 0
+    s1 = x[1] % 2
+    return s0 or (z0 and s1)
+
+
+def sswu(u):
+    zu2 = f2mul(Z_SSWU, f2sq(u))
+    tv = f2add(f2sq(zu2), zu2)  # Z^2 u^4 + Z u^2
+    if tv == F2_ZERO:
+        x1 = f2mul(B_P, f2inv(f2mul(Z_SSWU, A_P)))  # B/(Z*A)
+    else:
+        x1 = f2mul(
+            f2mul(f2neg(B_P), f2inv(A_P)), f2add(F2_ONE, f2inv(tv))
+        )  # (-B/A)(1 + 1/tv)
+    gx1 = f2add(f2mul(f2sq(x1), x1), f2add(f2mul(A_P, x1), B_P))
+    y1 = f2sqrt(gx1)
+    if y1 is not None:
+        x, y = x1, y1
+    else:
+        x2 = f2mul(zu2, x1)
+        gx2 = f2add(f2mul(f2sq(x2), x2), f2add(f2mul(A_P, x2), B_P))
+        y2 = f2sqrt(gx2)
+        assert y2 is not None
+        x, y = x2, y2
+    if sgn0_fp2(u) != sgn0_fp2(y):
+        y = f2neg(y)
+    # on E2'
+    assert f2sub(
+        f2sq(y), f2add(f2mul(f2sq(x), x), f2add(f2mul(A_P, x), B_P))
+    ) == F2_ZERO
+    return (x, y)
+
+
+# 3-isogeny E2' -> E2 (RFC 9380 App. E.3 constants)
+def _h(c0, c1=0):
+    return (c0 % P, c1 % P)
+
+
+ISO_XNUM = [
+    _h(
+        0x05C759507E8E333EBB5B7A9A47D7ED8532C52D39FD3A042A88B58423C50AE15D5C2638E343D9C71C6238AAAAAAAA97D6,
+        0x05C759507E8E333EBB5B7A9A47D7ED8532C52D39FD3A042A88B58423C50AE15D5C2638E343D9C71C6238AAAAAAAA97D6,
+    ),
+    _h(
+        0,
+        0x11560BF17BAA99BC32126FCED787C88F984F87ADF7AE0C7F9A208C6B4F20A4181472AAA9CB8D555526A9FFFFFFFFC71A,
+    ),
+    _h(
+        0x11560BF17BAA99BC32126FCED787C88F984F87ADF7AE0C7F9A208C6B4F20A4181472AAA9CB8D555526A9FFFFFFFFC71E,
+        0x08AB05F8BDD54CDE190937E76BC3E447CC27C3D6FBD7063FCD104635A790520C0A395554E5C6AAAA9354FFFFFFFFE38D,
+    ),
+    _h(
+        0x171D6541FA38CCFAED6DEA691F5FB614CB14B4E7F4E810AA22D6108F142B85757098E38D0F671C7188E2AAAAAAAA5ED1,
+        0,
+    ),
+]
+ISO_XDEN = [
+    _h(
+        0,
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFAA63,
+    ),
+    _h(
+        0xC,
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFAA9F,
+    ),
+    _h(1, 0),  # monic x^2
+]
+ISO_YNUM = [
+    _h(
+        0x1530477C7AB4113B59A4C18B076D11930F7DA5D4A07F649BF54439D87D27E500FC8C25EBF8C92F6812CFC71C71C6D706,
+        0x1530477C7AB4113B59A4C18B076D11930F7DA5D4A07F649BF54439D87D27E500FC8C25EBF8C92F6812CFC71C71C6D706,
+    ),
+    _h(
+        0,
+        0x05C759507E8E333EBB5B7A9A47D7ED8532C52D39FD3A042A88B58423C50AE15D5C2638E343D9C71C6238AAAAAAAA97BE,
+    ),
+    _h(
+        0x11560BF17BAA99BC32126FCED787C88F984F87ADF7AE0C7F9A208C6B4F20A4181472AAA9CB8D555526A9FFFFFFFFC71C,
+        0x08AB05F8BDD54CDE190937E76BC3E447CC27C3D6FBD7063FCD104635A790520C0A395554E5C6AAAA9354FFFFFFFFE38F,
+    ),
+    _h(
+        0x124C9AD43B6CF79BFBF7043DE3811AD0761B0F37A1E26286B0E977C69AA274524E79097A56DC4BD9E1B371C71C718B10,
+        0,
+    ),
+]
+ISO_YDEN = [
+    _h(
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFA8FB,
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFA8FB,
+    ),
+    _h(
+        0,
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFA9D3,
+    ),
+    _h(
+        0x12,
+        0x1A0111EA397FE69A4B1BA7B6434BACD764774B84F38512BF6730D2A0F6B0F6241EABFFFEB153FFFFB9FEFFFFFFFFAA99,
+    ),
+    _h(1, 0),  # monic x^3
+]
+
+
+def iso_map(p):
+    x, y = p
+
+    def horner(coeffs):
+        acc = F2_ZERO
+        for c in reversed(coeffs):
+            acc = f2add(f2mul(acc, x), c)
+        return acc
+
+    xn, xd = horner(ISO_XNUM), horner(ISO_XDEN)
+    yn, yd = horner(ISO_YNUM), horner(ISO_YDEN)
+    X = f2mul(xn, f2inv(xd))
+    Y = f2mul(y, f2mul(yn, f2inv(yd)))
+    q = (X, Y)
+    assert on_g2(q), "isogeny output not on E2 — iso constant typo"
+    return q
+
+
+# psi endomorphism constants — determined numerically, then asserted.
+def _find_psi():
+    cands = []
+    e3 = (P - 1) // 3
+    e2 = (P - 1) // 2
+    for cx in [f2pow(XI, e3), f2inv(f2pow(XI, e3))]:
+        for cy in [f2pow(XI, e2), f2inv(f2pow(XI, e2))]:
+            cands.append((cx, cy))
+    g = (G2X, G2Y)
+    lam = X_PARAM % R
+    want = g2_mul(lam, g)
+    for cx, cy in cands:
+        qq = (f2mul(cx, f2conj(G2X)), f2mul(cy, f2conj(G2Y)))
+        if on_g2(qq) and qq == want:
+            return cx, cy
+    raise AssertionError("psi constants not found")
+
+
+PSI_CX, PSI_CY = _find_psi()
+
+
+def psi(p):
+    if p is None:
+        return None
+    return (f2mul(PSI_CX, f2conj(p[0])), f2mul(PSI_CY, f2conj(p[1])))
+
+
+def clear_cofactor_g2(p):
+    """Budroni–Pintore: [x^2-x-1]P + [x-1]psi(P) + psi^2([2]P); equals the
+    RFC 9380 h_eff multiplication (asserted below on random points)."""
+    xp = g2_neg(g2_mul(H_EFF_ABS, p))  # [x]P, x negative
+    xxp = g2_neg(g2_mul(H_EFF_ABS, xp))  # [x^2]P
+    part1 = g2_add(g2_add(xxp, g2_neg(xp)), g2_neg(p))  # [x^2-x-1]P
+    part2 = psi(g2_add(xp, g2_neg(p)))  # [x-1]psi(P)
+    part3 = psi(psi(g2_add(p, p)))  # psi^2([2]P)
+    return g2_add(g2_add(part1, part2), part3)
+
+
+def hash_to_curve_g2(msg):
+    u0, u1 = hash_to_field_fp2(msg, 2)
+    q0 = iso_map(sswu(u0))
+    q1 = iso_map(sswu(u1))
+    q = clear_cofactor_g2(g2_add(q0, q1))
+    assert on_g2(q) and g2_mul(R, q) is None, "h2c output not in G2"
+    return q
+
+
+# ---------------------------------------------------------------- pairing
+# Fp12 = Fp2[w]/(w^6 - XI): list of 6 Fp2 coefficients.
+F12_ONE = [F2_ONE] + [F2_ZERO] * 5
+F12_ZERO = [F2_ZERO] * 6
+
+
+def f12mul(a, b):
+    acc = [F2_ZERO] * 11
+    for i in range(6):
+        if a[i] == F2_ZERO:
+            continue
+        for j in range(6):
+            if b[j] == F2_ZERO:
+                continue
+            acc[i + j] = f2add(acc[i + j], f2mul(a[i], b[j]))
+    out = acc[:6]
+    for k in range(6, 11):
+        out[k - 6] = f2add(out[k - 6], f2mul(acc[k], XI))
+    return out
+
+
+def f12conj6(f):
+    """f^(p^6): w -> w * XI^((p^6-1)/6); asserted below that the constant
+    is -1, i.e. odd coefficients negate."""
+    return [f[i] if i % 2 == 0 else f2neg(f[i]) for i in range(6)]
+
+
+_c6 = f2pow(XI, (P**6 - 1) // 6)
+assert _c6 == ((P - 1) % P, 0), "XI^((p^6-1)/6) != -1"
+
+
+def f12pow(a, e):
+    r = F12_ONE
+    while e:
+        if e & 1:
+            r = f12mul(r, a)
+        a = f12mul(a, a)
+        e >>= 1
+    return r
+
+
+def untwist(q):
+    """E'(Fp2) -> E(Fp12): (x,y) -> (x * xi^-1 * w^4, y * xi^-1 * w^3)."""
+    xi_inv = f2inv(XI)
+    X = list(F12_ZERO)
+    X[4] = f2mul(q[0], xi_inv)
+    Y = list(F12_ZERO)
+    Y[3] = f2mul(q[1], xi_inv)
+    return (X, Y)
+
+
+def f12inv_generic(a):
+    # product of the 5 nontrivial sigma-conjugates (w -> zeta6^i w), then
+    # reduce by the norm which lands in Fp2.
+    zeta6 = f2pow(XI, (P * P - 1) // 6)
+    assert f2pow(zeta6, 6) == F2_ONE and f2pow(zeta6, 3) != F2_ONE
+    g = F12_ONE
+    for i in range(1, 6):
+        zi = f2pow(zeta6, i)
+        conj = [f2mul(a[j], f2pow(zi, j)) for j in range(6)]
+        g = f12mul(g, conj)
+    n = f12mul(a, g)
+    assert all(n[j] == F2_ZERO for j in range(1, 6)), "norm not in Fp2"
+    ninv = f2inv(n[0])
+    return [f2mul(g[j], ninv) for j in range(6)]
+
+
+def miller_loop(pairs):
+    """prod over (P in G1 affine, Q in G2 affine) of f_{|x|,Q'}(P'), then
+    conjugated (x<0). Verticals omitted (even embedding degree: they lie in
+    the Fp6 subfield Fp2[w^2] and die in the final exponentiation)."""
+    f = F12_ONE
+    state = []
+    for (pp, qq) in pairs:
+        if pp is None or qq is None:
+            continue  # e(O, .) = e(., O) = 1
+        Q = untwist(qq)
+        state.append([Q, Q, (pp[0], pp[1])])  # [T, Q, P]
+    bits = bin(H_EFF_ABS)[3:]  # skip MSB
+    for b in bits:
+        f = f12mul(f, f)
+        for st in state:
+            T, Q, pp = st
+            xp, yp = pp
+            # doubling line at T evaluated at P: yP - yT - lam(xP - xT)
+            lam = f12mul(
+                f12mul(f12mul(T[0], T[0]), [f2smul(F2_ONE, 3)] + [F2_ZERO] * 5),
+                f12inv_generic([f2smul(c, 2) for c in T[1]]),
+            )
+            l = _line_eval(lam, T, xp, yp)
+            f = f12mul(f, l)
+            st[0] = _ec12_add(T, T, lam)
+        if b == "1":
+            for st in state:
+                T, Q, pp = st
+                xp, yp = pp
+                lam = f12mul(
+                    _f12sub(Q[1], T[1]), f12inv_generic(_f12sub(Q[0], T[0]))
+                )
+                l = _line_eval(lam, T, xp, yp)
+                f = f12mul(f, l)
+                st[0] = _ec12_add(T, Q, lam)
+    return f12conj6(f)  # x < 0
+
+
+def _f12sub(a, b):
+    return [f2sub(a[i], b[i]) for i in range(6)]
+
+
+def _f12add(a, b):
+    return [f2add(a[i], b[i]) for i in range(6)]
+
+
+def _line_eval(lam, T, xp, yp):
+    # l = yP - yT - lam*(xP - xT); xp,yp are base-field ints
+    xp12 = [(xp, 0)] + [F2_ZERO] * 5
+    yp12 = [(yp, 0)] + [F2_ZERO] * 5
+    return _f12sub(_f12sub(yp12, T[1]), f12mul(lam, _f12sub(xp12, T[0])))
+
+
+def _ec12_add(A, Bp, lam):
+    # affine add with precomputed lambda (works for double too)
+    x3 = _f12sub(_f12sub(f12mul(lam, lam), A[0]), Bp[0])
+    y3 = _f12sub(f12mul(lam, _f12sub(A[0], x3)), A[1])
+    return (x3, y3)
+
+
+FINAL_EXP = (P**12 - 1) // R
+
+
+def pairing_prod_is_one(pairs):
+    f = miller_loop(pairs)
+    return f12pow(f, FINAL_EXP) == F12_ONE
+
+
+def pairing(pp, qq):
+    return f12pow(miller_loop([(pp, qq)]), FINAL_EXP)
+
+
+# ---------------------------------------------------------------- BLS scheme
+G1G = (G1X, G1Y)
+G2G = (G2X, G2Y)
+
+
+def keygen_interop(index):
+    """eth2_interop_keypairs/src/lib.rs:40-55"""
+    pre = index.to_bytes(8, "little") + b"\x00" * 24
+    h = hashlib.sha256(pre).digest()
+    return int.from_bytes(h, "little") % R
+
+
+def sk_to_pk(sk):
+    return g1_mul(sk, G1G)
+
+
+def sign(sk, msg):
+    return g2_mul(sk, hash_to_curve_g2(msg))
+
+
+def verify(pk, msg, sig):
+    # e(pk, H(m)) == e(g1, sig)  <=>  e(-pk,H(m))*e(g1,sig) == 1
+    return pairing_prod_is_one([(g1_neg(pk), hash_to_curve_g2(msg)), (G1G, sig)])
+
+
+def verify_sets(sets, rands):
+    """blst.rs:37-119 batch equation. sets: list of (msg bytes, sig point or
+    None, [pk points]); rands: 64-bit nonzero ints."""
+    if not sets:
+        return False
+    pairs = []
+    sig_acc = None
+    for (msg, sig, pks), r in zip(sets, rands):
+        # NB: sig=None here is the point at INFINITY (a valid subgroup
+        # element — the batch equation then fails naturally); the EMPTY
+        # (all-zero) serialization is rejected by the host layer before this
+        # math, per blst.rs:80-83.
+        if g2_mul(R, sig) is not None:
+            return False  # subgroup check (blst.rs:73-77)
+        if not pks:
+            return False  # blst.rs:86-89
+        apk = None
+        for pk in pks:
+            apk = g1_add(apk, pk)
+        if apk is None:
+            return False  # aggregate pk at infinity
+        pairs.append((g1_mul(r, apk), hash_to_curve_g2(msg)))
+        sig_acc = g2_add(sig_acc, g2_mul(r, sig))
+    pairs.append((g1_neg(G1G), sig_acc))
+    return pairing_prod_is_one(pairs)
+
+
+# ---------------------------------------------------------------- validation
+# Reference interop vectors — test DATA from
+# /root/reference/common/eth2_interop_keypairs/tests/generation.rs:10-45
+REF_PRIVKEYS = [
+    "16808672146709759238327133555736750089977066230599028589193936481731504400486",
+    "37006103240406073079686739739280712467525465637222501547219594975923976982528",
+    "22330876536127119444572216874798222843352868708084730796787004036811744442455",
+    "17048462031355941381150076874414096388968985457797372268770826099852902060945",
+    "28647806952216650698330424381872693846361470773871570637461872359310549743691",
+    "2416304019107052589452838695606585506736351107897780798170812672519914514344",
+    "7300215445567548136411883691093515822872548648751398235557229381530420545683",
+    "26495790445032093722332687600112008700915252495659977774957922313678954054133",
+    "2908643403277969554503670470854573663206729491025062456164283925661321952518",
+    "19554639423851580804889717218680781396599791537051606512605582393920758869044",
+]
+REF_PUBKEYS_B64 = [
+    "qZp27XeW974i1bfoXe63xWd+iOUR4LM3YY+MTrYTSbS/LRU/ZJ97UzWf6LlKOORM",
+    "uJvrxpl2lyajGMjplxvTFxKXxhrqSmV4p6T5S1R9y6W6wWqJEItrah/jaV0ah0oL",
+    "o6MrD4tN24PxoKhT2B3XJd/ld9T0w9uOzlLOKwJuyoSBXBp+jpKk3j11VzO/fkqb",
+    "iMFB33fNnY16cadcgmxBqcnwPG7hsYDz54UvaigAmd7TUbWNZuZTr45CgWpNj1Mu",
+    "gSg7eiDhykYOvZu9dwBdVXNwyrsfmkT1MMTExmIw9nX434tMKBiFGqfXeoDKWkpe",
+    "qwvdoPhfhC9DG+rM8SUL8f17pRtBAP1kNktkAf2oW7AGmz5xW1iBloTn/AsQpyo0",
+    "mXfxyLcxqNVVgUa/uGyuomQ088WHi1ib8oCkLJFZ5wDp3w5AhilsILAR0ueMJ9Nz",
+    "qNTHwneVpyWWExfvWVOnAy7W2Dc524sOinI1PRuLRDlCf376LInKoDzJ8o+Muris",
+    "ptMQ27+rmiJFD1mZP4ekzl22Ij87Xx8w0sTscYki1ADgs8d0HejlmWD3JBGg7hCn",
+    "mJNBPAAoOj+e2f2YRd2hzqOCKNIlZ/lUHczDV+VKLWpuIEEDySVky8BfSQWsfEk6",
+]
+
+
+def validate():
+    print("validating against reference interop vectors ...")
+    for i in range(10):
+        sk = keygen_interop(i)
+        assert str(sk) == REF_PRIVKEYS[i], f"privkey {i} mismatch"
+        pk = sk_to_pk(sk)
+        want = base64.b64decode(REF_PUBKEYS_B64[i])
+        got = g1_compress(pk)
+        assert got == want, f"pubkey {i} mismatch: {got.hex()} vs {want.hex()}"
+    print("  interop keygen + G1 mult + compression: OK (10/10)")
+
+    # round-trip serialization
+    for i in range(3):
+        sk = keygen_interop(i)
+        pk = sk_to_pk(sk)
+        assert g1_decompress(g1_compress(pk)) == pk
+        sig = sign(sk, bytes([i]) * 32)
+        assert g2_decompress(g2_compress(sig)) == sig
+    print("  serialization round-trips: OK")
+
+    # clear_cofactor (psi form) == h_eff scalar mult on a random E2 point
+    H_EFF_RFC = int(
+        "bc69f08f2ee75b3584c6a0ea91b352888e2a8e9145ad7689986ff031508ffe13"
+        "29c2f178731db956d82bf015d1212b02ec0ec69d7477c1ae954cbc06689f6a35"
+        "9894c0adebbf6b4e8020005aaa95551",
+        16,
+    )
+    rng_x = 5
+    while True:
+        gx = f2add(f2mul(f2sq((rng_x, 3)), (rng_x, 3)), B2)
+        y = f2sqrt(gx)
+        if y is not None:
+            pt = ((rng_x, 3), y)
+            break
+        rng_x += 1
+    a = clear_cofactor_g2(pt)
+    b = g2_mul(H_EFF_RFC, pt)
+    assert a == b, "psi clear_cofactor != h_eff mult (constant memory error)"
+    print("  clear_cofactor psi-form == RFC h_eff: OK")
+
+    # pairing bilinearity
+    e_ab = pairing(g1_mul(5, G1G), g2_mul(7, G2G))
+    e_1 = pairing(G1G, G2G)
+    assert e_ab == f12pow(e_1, 35), "bilinearity failed"
+    assert e_1 != F12_ONE, "degenerate pairing"
+    print("  pairing bilinearity e(5G1,7G2)=e(G1,G2)^35: OK")
+
+    # sign/verify
+    sk = keygen_interop(0)
+    msg = b"\x11" * 32
+    sig = sign(sk, msg)
+    assert verify(sk_to_pk(sk), msg, sig)
+    assert not verify(sk_to_pk(sk), b"\x22" * 32, sig)
+    print("  sign/verify: OK")
+
+
+# ---------------------------------------------------------------- emission
+def limbs(v, n=6):
+    return [(v >> (64 * i)) & 0xFFFFFFFFFFFFFFFF for i in range(n)]
+
+
+def c_limbs(name, v, n=6):
+    ls = ", ".join(f"0x{w:016x}ULL" for w in limbs(v, n))
+    return f"static const uint64_t {name}[{n}] = {{{ls}}};\n"
+
+
+def c_fp2(name, v):
+    return c_limbs(name + "_C0", v[0]) + c_limbs(name + "_C1", v[1])
+
+
+def emit_consts(path):
+    R_MONT = (1 << 384) % P
+    R2 = R_MONT * R_MONT % P
+    N0 = (-pow(P, -1, 1 << 64)) % (1 << 64)
+    D_HARD = (P**4 - P**2 + 1) // R
+    FW1 = f2pow(XI, (P - 1) // 6)  # w^p = FW1 * w
+    out = []
+    out.append("/* GENERATED by tests/golden/gen_bls_fixtures.py — do not edit.\n")
+    out.append(" * All constants validated by numeric asserts in the generator\n")
+    out.append(" * (on-curve, order, psi==h_eff, bilinearity, interop vectors).\n")
+    out.append(" * Limbs are little-endian 64-bit, values in STANDARD (non-\n")
+    out.append(" * Montgomery) form unless noted. */\n")
+    out.append("#ifndef M3X_BLS_CONSTS_H\n#define M3X_BLS_CONSTS_H\n")
+    out.append("#include <stdint.h>\n")
+    out.append(c_limbs("BLS_P", P))
+    out.append(c_limbs("BLS_R2", R2))  # Montgomery R^2 mod p
+    out.append(f"static const uint64_t BLS_N0 = 0x{N0:016x}ULL;\n")
+    out.append(c_limbs("BLS_ORDER", R, 4))
+    out.append(f"#define BLS_X_ABS 0x{H_EFF_ABS:016x}ULL\n")
+    out.append("#define BLS_X_NEGATIVE 1\n")
+    out.append(c_limbs("BLS_G1X", G1X))
+    out.append(c_limbs("BLS_G1Y", G1Y))
+    out.append(c_fp2("BLS_G2X", G2X))
+    out.append(c_fp2("BLS_G2Y", G2Y))
+    out.append(c_fp2("SSWU_A", A_P))
+    out.append(c_fp2("SSWU_B", B_P))
+    out.append(c_fp2("SSWU_Z", Z_SSWU))
+    for nm, arr in [
+        ("ISO_XNUM", ISO_XNUM),
+        ("ISO_XDEN", ISO_XDEN),
+        ("ISO_YNUM", ISO_YNUM),
+        ("ISO_YDEN", ISO_YDEN),
+    ]:
+        for i, c in enumerate(arr):
+            out.append(c_fp2(f"{nm}{i}", c))
+        out.append(f"#define {nm}_N {len(arr)}\n")
+    out.append(c_fp2("PSI_CX", PSI_CX))
+    out.append(c_fp2("PSI_CY", PSI_CY))
+    out.append(c_fp2("FROB_W1", FW1))
+    # hard-part exponent d = (p^4 - p^2 + 1)/r, little-endian u64 limbs
+    nl = (D_HARD.bit_length() + 63) // 64
+    out.append(c_limbs("FINAL_EXP_D", D_HARD, nl))
+    out.append(f"#define FINAL_EXP_D_LIMBS {nl}\n")
+    out.append("#endif\n")
+    with open(path, "w") as f:
+        f.write("".join(out))
+    print(f"wrote {path}")
+
+
+def emit_fixtures(path):
+    import random
+
+    rng = random.Random(0xC0FFEE)
+    fx = {"comment": "generated by gen_bls_fixtures.py; pins the C oracle"}
+    fx["interop"] = []
+    for i in range(10):
+        sk = keygen_interop(i)
+        fx["interop"].append(
+            {
+                "index": i,
+                "sk_be_hex": sk.to_bytes(32, "big").hex(),
+                "pk_compressed_hex": g1_compress(sk_to_pk(sk)).hex(),
+                "pk_uncompressed_hex": g1_uncompressed(sk_to_pk(sk)).hex(),
+            }
+        )
+    fx["h2c"] = []
+    for msg in [b"\x00" * 32, b"\xab" * 32, hashlib.sha256(b"m3x").digest()]:
+        q = hash_to_curve_g2(msg)
+        fx["h2c"].append(
+            {"msg_hex": msg.hex(), "g2_uncompressed_hex": g2_uncompressed(q).hex()}
+        )
+    fx["signatures"] = []
+    for i in range(4):
+        sk = keygen_interop(i)
+        msg = hashlib.sha256(b"sigmsg" + bytes([i])).digest()
+        sig = sign(sk, msg)
+        fx["signatures"].append(
+            {
+                "index": i,
+                "msg_hex": msg.hex(),
+                "sig_compressed_hex": g2_compress(sig).hex(),
+                "sig_uncompressed_hex": g2_uncompressed(sig).hex(),
+            }
+        )
+    # GT fixture: e(G1,G2) serialized as 12 x 48B (c0..c5, each c0||c1 BE)
+    e1 = pairing(G1G, G2G)
+
+    def gt_hex(f):
+        b = b""
+        for c in f:
+            b += c[0].to_bytes(48, "big") + c[1].to_bytes(48, "big")
+        return b.hex()
+
+    fx["gt"] = {
+        "e_g1_g2_hex": gt_hex(e1),
+        "e_5g1_7g2_hex": gt_hex(pairing(g1_mul(5, G1G), g2_mul(7, G2G))),
+    }
+    # batch-verify cases (msgs 32B, sigs compressed, pk sets by interop index)
+    cases = []
+
+    def mk_case(name, set_specs, tamper=None):
+        sets_py = []
+        sets_js = []
+        rands = [rng.getrandbits(64) | 1 for _ in set_specs]
+        for (idxs, msg) in set_specs:
+            sks = [keygen_interop(i) for i in idxs]
+            agg = None
+            for sk in sks:
+                agg = g2_add(agg, sign(sk, msg))
+            sets_py.append((msg, agg, [sk_to_pk(sk) for sk in sks]))
+            sets_js.append(
+                {
+                    "msg_hex": msg.hex(),
+                    "sig_compressed_hex": g2_compress(agg).hex(),
+                    "pks_uncompressed_hex": [
+                        g1_uncompressed(sk_to_pk(sk)).hex() for sk in sks
+                    ],
+                }
+            )
+        if tamper is not None:
+            sets_py, sets_js = tamper(sets_py, sets_js)
+        verdict = verify_sets(sets_py, rands)
+        cases.append(
+            {
+                "name": name,
+                "rands": [str(r) for r in rands],
+                "sets": sets_js,
+                "verdict": verdict,
+            }
+        )
+
+    m = lambda s: hashlib.sha256(s).digest()
+    mk_case("single_k1_valid", [([0], m(b"a"))])
+    mk_case("three_valid_mixed_k", [([0], m(b"a")), ([1, 2, 3], m(b"b")), ([4, 5], m(b"c"))])
+
+    def tamper_msg(py, js):
+        bad = m(b"evil")
+        py[0] = (bad, py[0][1], py[0][2])
+        js[0]["msg_hex"] = bad.hex()
+        return py, js
+
+    mk_case("bad_message", [([0], m(b"a")), ([1], m(b"b"))], tamper_msg)
+
+    def tamper_sig_infinity(py, js):
+        py[1] = (py[1][0], None, py[1][2])
+        js[1]["sig_compressed_hex"] = g2_compress(None).hex()
+        return py, js
+
+    mk_case(
+        "infinity_sig", [([0], m(b"a")), ([1], m(b"b"))], tamper_sig_infinity
+    )
+
+    def tamper_wrong_key(py, js):
+        pk9 = sk_to_pk(keygen_interop(9))
+        py[0] = (py[0][0], py[0][1], [pk9])
+        js[0]["pks_uncompressed_hex"] = [g1_uncompressed(pk9).hex()]
+        return py, js
+
+    mk_case("wrong_pubkey", [([0], m(b"a"))], tamper_wrong_key)
+    mk_case(
+        "ten_sets_valid",
+        [([i], m(bytes([i]) * 3)) for i in range(10)],
+    )
+    fx["batch_cases"] = cases
+    with open(path, "w") as f:
+        json.dump(fx, f, indent=1)
+    print(f"wrote {path}")
+
+
+if __name__ == "__main__":
+    here = os.path.dirname(os.path.abspath(__file__))
+    repo = os.path.dirname(os.path.dirname(here))
+    validate()
+    emit_consts(os.path.join(repo, "oracle", "bls_consts.h"))
+    emit_fixtures(os.path.join(here, "bls_fixtures.json"))
+    print("all good")
