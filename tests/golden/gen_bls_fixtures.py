@@ -853,6 +853,9 @@ def emit_consts(path):
     out.append(c_fp2("PSI_CX", PSI_CX))
     out.append(c_fp2("PSI_CY", PSI_CY))
     out.append(c_fp2("FROB_W1", FW1))
+    zeta6 = f2pow(XI, (P * P - 1) // 6)  # primitive 6th root of unity in Fp2
+    assert f2pow(zeta6, 6) == F2_ONE and f2pow(zeta6, 3) != F2_ONE
+    out.append(c_fp2("ZETA6", zeta6))
     # hard-part exponent d = (p^4 - p^2 + 1)/r, little-endian u64 limbs
     nl = (D_HARD.bit_length() + 63) // 64
     out.append(c_limbs("FINAL_EXP_D", D_HARD, nl))
@@ -910,6 +913,12 @@ def emit_fixtures(path):
     fx["gt"] = {
         "e_g1_g2_hex": gt_hex(e1),
         "e_5g1_7g2_hex": gt_hex(pairing(g1_mul(5, G1G), g2_mul(7, G2G))),
+    }
+    fx["generators"] = {
+        "g1_uncompressed_hex": g1_uncompressed(G1G).hex(),
+        "g2_uncompressed_hex": g2_uncompressed(G2G).hex(),
+        "g1_times5_uncompressed_hex": g1_uncompressed(g1_mul(5, G1G)).hex(),
+        "g2_times7_uncompressed_hex": g2_uncompressed(g2_mul(7, G2G)).hex(),
     }
     # batch-verify cases (msgs 32B, sigs compressed, pk sets by interop index)
     cases = []
