@@ -1,0 +1,204 @@
+"""ctypes binding to libm3x_consensus.so (the C-ABI in include/m3x_consensus.h).
+
+Product path only — no CPU fallback: if a GPU is present and the library is
+missing or fails to load, every call raises. The oracle is never imported
+here (see DESIGN.md §Parity strategy)."""
+import ctypes
+import os
+from pathlib import Path
+
+_PKG = Path(__file__).resolve().parent
+_LIB_PATH = _PKG / "libm3x_consensus.so"
+
+_lib = None
+_load_error = None
+
+
+def _bind(lib):
+    lib.m3x_abi_version.restype = ctypes.c_int32
+    lib.m3x_ctx_create.argtypes = [ctypes.POINTER(ctypes.c_void_p), ctypes.c_int32]
+    lib.m3x_ctx_create.restype = ctypes.c_int32
+    lib.m3x_ctx_destroy.argtypes = [ctypes.c_void_p]
+    lib.m3x_dev_alloc.argtypes = [
+        ctypes.c_void_p,
+        ctypes.c_uint64,
+        ctypes.POINTER(ctypes.c_void_p),
+    ]
+    lib.m3x_dev_alloc.restype = ctypes.c_int32
+    lib.m3x_dev_free.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.m3x_dev_free.restype = ctypes.c_int32
+    lib.m3x_h2d.argtypes = [
+        ctypes.c_void_p,
+        ctypes.c_void_p,
+        ctypes.c_char_p,
+        ctypes.c_uint64,
+    ]
+    lib.m3x_h2d.restype = ctypes.c_int32
+    lib.m3x_d2h.argtypes = [
+        ctypes.c_void_p,
+        ctypes.c_void_p,
+        ctypes.c_void_p,
+        ctypes.c_uint64,
+    ]
+    lib.m3x_d2h.restype = ctypes.c_int32
+    for name, argtypes in [
+        (
+            "m3x_merkleize_validators",
+            [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint64, ctypes.c_char_p],
+        ),
+        (
+            "m3x_merkleize_validators_dev",
+            [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64, ctypes.c_char_p],
+        ),
+        (
+            "m3x_merkleize_chunks",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_uint64,
+                ctypes.c_uint32,
+                ctypes.c_int64,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
+            "m3x_merkleize_chunks_dev",
+            [
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_uint64,
+                ctypes.c_uint32,
+                ctypes.c_int64,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
+            "m3x_validator_subtree_root_dev",
+            [
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_uint64,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
+            "m3x_bls_pk_decompress",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_uint64,
+                ctypes.c_char_p,
+                ctypes.POINTER(ctypes.c_int32),
+            ],
+        ),
+        (
+            "m3x_bls_verify_sets",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_char_p,
+                ctypes.c_char_p,
+                ctypes.POINTER(ctypes.c_uint32),
+                ctypes.POINTER(ctypes.c_uint64),
+                ctypes.c_uint64,
+            ],
+        ),
+        (
+            "m3x_bls_verify_sets_dev",
+            [
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_uint64,
+            ],
+        ),
+    ]:
+        try:
+            fn = getattr(lib, name)
+        except AttributeError:
+            continue
+        fn.argtypes = argtypes
+        fn.restype = ctypes.c_int32
+    return lib
+
+
+def load():
+    """Load (and cache) the native library; raises RuntimeError if absent."""
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    if not _LIB_PATH.exists():
+        raise RuntimeError(
+            f"m3x_consensus native library missing: {_LIB_PATH}. "
+            "Build it with `python -c 'import __graft_entry__; "
+            "__graft_entry__.build()'` — there is no CPU fallback."
+        )
+    try:
+        _lib = _bind(ctypes.CDLL(str(_LIB_PATH)))
+    except OSError as e:
+        _load_error = e
+        raise RuntimeError(f"failed to load {_LIB_PATH}: {e}") from e
+    return _lib
+
+
+class Ctx:
+    """One GPU context (device + stream + scratch)."""
+
+    def __init__(self, device: int = 0):
+        self._lib = load()
+        self._h = ctypes.c_void_p()
+        rc = self._lib.m3x_ctx_create(ctypes.byref(self._h), device)
+        if rc != 0:
+            raise RuntimeError(
+                f"m3x_ctx_create failed (rc={rc}) — is an MI355X visible?"
+            )
+
+    def close(self):
+        if self._h:
+            self._lib.m3x_ctx_destroy(self._h)
+            self._h = ctypes.c_void_p()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    @property
+    def handle(self):
+        return self._h
+
+    # ---- device buffers ----
+    def alloc(self, nbytes: int) -> ctypes.c_void_p:
+        p = ctypes.c_void_p()
+        rc = self._lib.m3x_dev_alloc(self._h, nbytes, ctypes.byref(p))
+        if rc != 0:
+            raise RuntimeError(f"m3x_dev_alloc({nbytes}) rc={rc}")
+        return p
+
+    def free(self, p):
+        self._lib.m3x_dev_free(self._h, p)
+
+    def h2d(self, dev, host_bytes: bytes):
+        rc = self._lib.m3x_h2d(self._h, dev, host_bytes, len(host_bytes))
+        if rc != 0:
+            raise RuntimeError(f"m3x_h2d rc={rc}")
+
+    def upload(self, host_bytes: bytes):
+        p = self.alloc(max(len(host_bytes), 4))
+        self.h2d(p, host_bytes)
+        return p
+
+
+_default_ctx = None
+
+
+def default_ctx() -> Ctx:
+    global _default_ctx
+    if _default_ctx is None:
+        _default_ctx = Ctx(int(os.environ.get("M3X_DEVICE", "0")))
+    return _default_ctx

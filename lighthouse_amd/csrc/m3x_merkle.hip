@@ -1,0 +1,462 @@
+// MI355X-native SSZ merkleization (hot path #2) + context/buffer C-ABI.
+//
+// Replaces BeaconState::update_tree_hash_cache's SHA256 merkleize
+// (consensus/types/src/beacon_state.rs:2031-2046; math restated from
+// merkle_proof/src/lib.rs:9-14,68-100, mix_in_length from
+// deposit_data_tree.rs:26-38, Validator layout from validator.rs:25-35,
+// 2^40 registry limit from eth_spec.rs:404). Designed CDNA4-first: one
+// validator / one two-to-one node per lane (wave64), the padding-block
+// compression constant-folded, multi-level LDS reduction per workgroup,
+// grids ≫256 blocks to fill 8 XCDs. Integer VALU workload — no MFMA, HBM
+// traffic ≈96B per node (see DESIGN.md roofline).
+#include "sha256.hh"
+#include "m3x_ctx.hh"
+#include "../../include/m3x_consensus.h"
+#include <vector>
+
+using namespace m3x;
+
+// ------------------------------------------------------------------ kernels
+
+// single-thread: Z[0]=0^32, Z[i]=H(Z[i-1]||Z[i-1]) (merkle_proof ZERO_NODES)
+__global__ void k_zero_ladder(uint8_t *zeros /*65*32*/) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  uint32_t z[8];
+#pragma unroll
+  for (int i = 0; i < 8; i++) z[i] = 0;
+  for (int j = 0; j < 32; j++) zeros[j] = 0;
+  for (int lvl = 1; lvl <= 64; lvl++) {
+    uint32_t o[8];
+    sha256_node(z, z, o);
+#pragma unroll
+    for (int i = 0; i < 8; i++) z[i] = o[i];
+    uint8_t *dst = zeros + 32 * lvl;
+    for (int i = 0; i < 8; i++) {
+      dst[4 * i] = (uint8_t)(z[i] >> 24);
+      dst[4 * i + 1] = (uint8_t)(z[i] >> 16);
+      dst[4 * i + 2] = (uint8_t)(z[i] >> 8);
+      dst[4 * i + 3] = (uint8_t)z[i];
+    }
+  }
+}
+
+__device__ __forceinline__ uint32_t be_load_u8x4(const uint8_t *p) {
+  return ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) |
+         ((uint32_t)p[2] << 8) | p[3];
+}
+
+// Validator hash_tree_root leaves: one lane = one validator (121B SSZ ->
+// 32B root, 8 two-to-one hashes: pubkey root + 7 field-tree nodes).
+// The block cooperatively stages its 256*121 contiguous bytes through LDS
+// with coalesced u32 loads, then each lane assembles its record.
+__global__ __launch_bounds__(256) void k_validator_leaves(
+    const uint8_t *__restrict__ ssz, uint64_t n, uint8_t *__restrict__ out) {
+  __shared__ uint8_t lds[256 * 121 + 4];
+  uint64_t base_rec = (uint64_t)blockIdx.x * 256;
+  const uint32_t block_bytes = 256 * 121; // 30976, divisible by 4
+  uint64_t gbase = base_rec * 121;
+  uint64_t total = n * 121;
+  // coalesced staging (u32 granularity; source is 4-byte aligned per block)
+  for (uint32_t i = threadIdx.x; i < block_bytes / 4; i += 256) {
+    uint64_t goff = gbase + 4ull * i;
+    uint32_t v = 0;
+    if (goff + 4 <= total) {
+      v = *reinterpret_cast<const uint32_t *>(ssz + goff);
+    } else if (goff < total) {
+      // ragged tail of the whole array
+      uint32_t b0 = ssz[goff];
+      uint32_t b1 = (goff + 1 < total) ? ssz[goff + 1] : 0;
+      uint32_t b2 = (goff + 2 < total) ? ssz[goff + 2] : 0;
+      v = b0 | (b1 << 8) | (b2 << 16);
+    }
+    *reinterpret_cast<uint32_t *>(lds + 4 * i) = v;
+  }
+  __syncthreads();
+  uint64_t rec = base_rec + threadIdx.x;
+  if (rec >= n) return;
+  const uint8_t *v = lds + 121 * threadIdx.x;
+  // field chunks (big-endian words for the hasher); layout validator.rs:25-35
+  uint32_t c[8][8];
+#pragma unroll
+  for (int i = 0; i < 8; i++)
+#pragma unroll
+    for (int j = 0; j < 8; j++) c[i][j] = 0;
+  // c0 = H(pk[0:32] || pk[32:48] + 16 zero bytes)
+  {
+    uint32_t l[8], r[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++) l[i] = be_load_u8x4(v + 4 * i);
+#pragma unroll
+    for (int i = 0; i < 4; i++) r[i] = be_load_u8x4(v + 32 + 4 * i);
+#pragma unroll
+    for (int i = 4; i < 8; i++) r[i] = 0;
+    sha256_node(l, r, c[0]);
+  }
+#pragma unroll
+  for (int i = 0; i < 8; i++) c[1][i] = be_load_u8x4(v + 48 + 4 * i); // wc
+  c[2][0] = be_load_u8x4(v + 80); // effective_balance LE bytes as-is
+  c[2][1] = be_load_u8x4(v + 84);
+  c[3][0] = (uint32_t)v[88] << 24; // slashed bool in byte 0 of the chunk
+  c[4][0] = be_load_u8x4(v + 89);
+  c[4][1] = be_load_u8x4(v + 93);
+  c[5][0] = be_load_u8x4(v + 97);
+  c[5][1] = be_load_u8x4(v + 101);
+  c[6][0] = be_load_u8x4(v + 105);
+  c[6][1] = be_load_u8x4(v + 109);
+  c[7][0] = be_load_u8x4(v + 113);
+  c[7][1] = be_load_u8x4(v + 117);
+  // 8-leaf tree: 4 + 2 + 1 hashes
+  uint32_t h01[8], h23[8], h45[8], h67[8], ha[8], hb[8], root[8];
+  sha256_node(c[0], c[1], h01);
+  sha256_node(c[2], c[3], h23);
+  sha256_node(c[4], c[5], h45);
+  sha256_node(c[6], c[7], h67);
+  sha256_node(h01, h23, ha);
+  sha256_node(h45, h67, hb);
+  sha256_node(ha, hb, root);
+  uint8_t *dst = out + 32 * rec;
+  uint32_t *dst32 = reinterpret_cast<uint32_t *>(dst);
+#pragma unroll
+  for (int i = 0; i < 8; i++) dst32[i] = __builtin_bswap32(root[i]);
+}
+
+// Multi-level reduction: each block folds 2^levels consecutive input nodes
+// (levels <= 9) into one output node through LDS double buffers. Missing
+// trailing nodes are the zero ladder Z[base_level] (padding with ladder
+// values is exactly the right-sparse semantics of merkle_proof::create).
+__global__ __launch_bounds__(256) void k_reduce(
+    const uint8_t *__restrict__ in, uint64_t n_in, uint8_t *__restrict__ out,
+    uint32_t levels, uint32_t base_level, const uint8_t *__restrict__ zeros) {
+  __shared__ uint32_t A[512][8];
+  __shared__ uint32_t B[256][8];
+  const uint32_t span = 1u << levels; // <= 512
+  uint64_t base = (uint64_t)blockIdx.x * span;
+  uint32_t zw[8];
+#pragma unroll
+  for (int i = 0; i < 8; i++)
+    zw[i] = be_load_u8x4(zeros + 32 * base_level + 4 * i);
+  // load span nodes (each thread loads up to 2)
+  for (uint32_t s = threadIdx.x; s < span; s += 256) {
+    uint64_t idx = base + s;
+    if (idx < n_in) {
+      const uint32_t *src = reinterpret_cast<const uint32_t *>(in + 32 * idx);
+#pragma unroll
+      for (int i = 0; i < 8; i++) A[s][i] = __builtin_bswap32(src[i]);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; i++) A[s][i] = zw[i];
+    }
+  }
+  __syncthreads();
+  uint32_t cnt = span;
+  for (uint32_t l = 0; l < levels; l++) {
+    uint32_t half = cnt >> 1;
+    if (l % 2 == 0) { // A -> B
+      for (uint32_t t = threadIdx.x; t < half; t += 256)
+        sha256_node(A[2 * t], A[2 * t + 1], B[t]);
+    } else { // B -> A
+      for (uint32_t t = threadIdx.x; t < half; t += 256)
+        sha256_node(B[2 * t], B[2 * t + 1], A[t]);
+    }
+    cnt = half;
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const uint32_t *res = (levels % 2 == 1) ? B[0] : A[0];
+    uint32_t *dst = reinterpret_cast<uint32_t *>(out + 32 * blockIdx.x);
+#pragma unroll
+    for (int i = 0; i < 8; i++) dst[i] = __builtin_bswap32(res[i]);
+  }
+}
+
+// single-thread finalize: carry node from `from_level` to `to_depth` against
+// the zero ladder (node is always the LEFT child — right-sparse tree), then
+// optionally mix_in_length.
+__global__ void k_finalize(const uint8_t *__restrict__ node_in,
+                           uint32_t from_level, uint32_t to_depth,
+                           int64_t mix_len, const uint8_t *__restrict__ zeros,
+                           uint8_t *__restrict__ out32) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  uint32_t node[8];
+  const uint32_t *src = reinterpret_cast<const uint32_t *>(node_in);
+#pragma unroll
+  for (int i = 0; i < 8; i++) node[i] = __builtin_bswap32(src[i]);
+  for (uint32_t l = from_level; l < to_depth; l++) {
+    uint32_t z[8], o[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++) z[i] = be_load_u8x4(zeros + 32 * l + 4 * i);
+    sha256_node(node, z, o);
+#pragma unroll
+    for (int i = 0; i < 8; i++) node[i] = o[i];
+  }
+  if (mix_len >= 0) {
+    // length chunk: LE64(len) || zeros  (deposit_data_tree.rs:26-38)
+    uint64_t len = (uint64_t)mix_len;
+    uint32_t lc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    lc[0] = __builtin_bswap32((uint32_t)(len & 0xffffffffu));
+    lc[1] = __builtin_bswap32((uint32_t)(len >> 32));
+    uint32_t o[8];
+    sha256_node(node, lc, o);
+#pragma unroll
+    for (int i = 0; i < 8; i++) node[i] = o[i];
+  }
+  uint32_t *dst = reinterpret_cast<uint32_t *>(out32);
+#pragma unroll
+  for (int i = 0; i < 8; i++) dst[i] = __builtin_bswap32(node[i]);
+}
+
+// ---------------------------------------------------------------- host side
+
+namespace m3x {
+int ensure_scratch(m3x_ctx *ctx, uint8_t **buf, uint64_t *cur,
+                   uint64_t bytes) {
+  if (*cur >= bytes) return M3X_OK;
+  if (*buf) (void)hipFree(*buf);
+  *buf = nullptr;
+  *cur = 0;
+  if (hipMalloc(buf, bytes) != hipSuccess) return M3X_ERR_NOMEM;
+  *cur = bytes;
+  return M3X_OK;
+}
+} // namespace m3x
+
+extern "C" {
+
+int32_t m3x_abi_version(void) { return 1; }
+
+int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
+  auto *ctx = new m3x_ctx();
+  ctx->device = device;
+  if (hipSetDevice(device) != hipSuccess) {
+    delete ctx;
+    return M3X_ERR_HIP;
+  }
+  if (hipStreamCreate(&ctx->stream) != hipSuccess) {
+    delete ctx;
+    return M3X_ERR_HIP;
+  }
+  if (hipMalloc(&ctx->zeros_dev, 65 * 32) != hipSuccess) {
+    delete ctx;
+    return M3X_ERR_NOMEM;
+  }
+  hipLaunchKernelGGL(k_zero_ladder, dim3(1), dim3(64), 0, ctx->stream,
+                     ctx->zeros_dev);
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) {
+    delete ctx;
+    return M3X_ERR_HIP;
+  }
+  *out = ctx;
+  return M3X_OK;
+}
+
+void m3x_ctx_destroy(m3x_ctx *ctx) {
+  if (!ctx) return;
+  if (ctx->zeros_dev) (void)hipFree(ctx->zeros_dev);
+  if (ctx->scratch_a) (void)hipFree(ctx->scratch_a);
+  if (ctx->scratch_b) (void)hipFree(ctx->scratch_b);
+  if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
+  delete ctx;
+}
+
+int32_t m3x_dev_alloc(m3x_ctx *ctx, uint64_t bytes, void **dev_ptr) {
+  if (!ctx || !dev_ptr) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  M3X_HIP_CHECK(hipMalloc(dev_ptr, bytes));
+  return M3X_OK;
+}
+
+int32_t m3x_dev_free(m3x_ctx *ctx, void *dev_ptr) {
+  if (!ctx) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipFree(dev_ptr));
+  return M3X_OK;
+}
+
+int32_t m3x_h2d(m3x_ctx *ctx, void *dst_dev, const void *src_host,
+                uint64_t bytes) {
+  if (!ctx) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  M3X_HIP_CHECK(hipMemcpyAsync(dst_dev, src_host, bytes, hipMemcpyHostToDevice,
+                               ctx->stream));
+  M3X_HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return M3X_OK;
+}
+
+int32_t m3x_d2h(m3x_ctx *ctx, void *dst_host, const void *src_dev,
+                uint64_t bytes) {
+  if (!ctx) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  M3X_HIP_CHECK(hipMemcpyAsync(dst_host, src_dev, bytes, hipMemcpyDeviceToHost,
+                               ctx->stream));
+  M3X_HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return M3X_OK;
+}
+
+// Reduce an array of nodes to a single subtree root at depth `depth`,
+// writing the 32B root into out_dev (device). Returns applied levels.
+static int32_t reduce_to_root(m3x_ctx *ctx, const uint8_t *nodes_dev,
+                              uint64_t n, uint32_t depth, uint8_t *out_dev) {
+  // alternate between scratch_b halves for stage outputs
+  const uint8_t *cur = nodes_dev;
+  uint64_t m = n;
+  uint32_t level = 0;
+  uint8_t *ping = nullptr, *pong = nullptr;
+  uint64_t need = (n + 511) / 512 * 32 + 64;
+  int rc = ensure_scratch(ctx, &ctx->scratch_b, &ctx->scratch_b_bytes,
+                          2 * need);
+  if (rc != M3X_OK) return rc;
+  ping = ctx->scratch_b;
+  pong = ctx->scratch_b + need;
+  while (level < depth && m > 1) {
+    uint32_t levels = depth - level < 9 ? depth - level : 9;
+    uint64_t span = 1ull << levels;
+    uint64_t n_out = (m + span - 1) / span;
+    uint8_t *dst = (cur == ping) ? pong : ping;
+    hipLaunchKernelGGL(k_reduce, dim3((uint32_t)n_out), dim3(256), 0,
+                       ctx->stream, cur, m, dst, levels, level,
+                       ctx->zeros_dev);
+    cur = dst;
+    m = n_out;
+    level += levels;
+  }
+  // finalize: carry to full depth (and no mix here)
+  hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream, cur, level,
+                     depth, (int64_t)-1, ctx->zeros_dev, out_dev);
+  return M3X_OK;
+}
+
+int32_t m3x_merkleize_chunks_dev(m3x_ctx *ctx, const void *chunks_dev,
+                                 uint64_t n_chunks, uint32_t depth,
+                                 int64_t mix_len, uint8_t out_root[32]) {
+  if (!ctx) return M3X_ERR_ARG;
+  if (n_chunks > (1ull << depth) && depth < 63) return M3X_ERR_ARG;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *root_dev;
+  M3X_HIP_CHECK(hipMalloc(&root_dev, 64));
+  int32_t rc;
+  if (n_chunks == 0) {
+    // Z[depth] then optional mix
+    hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream,
+                       ctx->zeros_dev + 32 * depth, depth, depth, mix_len,
+                       ctx->zeros_dev, root_dev);
+  } else {
+    rc = reduce_to_root(ctx, (const uint8_t *)chunks_dev, n_chunks, depth,
+                        root_dev + 32);
+    if (rc != M3X_OK) {
+      (void)hipFree(root_dev);
+      return rc;
+    }
+    hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream,
+                       root_dev + 32, depth, depth, mix_len, ctx->zeros_dev,
+                       root_dev);
+  }
+  int32_t r2 = M3X_OK;
+  if (hipMemcpyAsync(out_root, root_dev, 32, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    r2 = M3X_ERR_HIP;
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
+  (void)hipFree(root_dev);
+  return r2;
+}
+
+int32_t m3x_merkleize_chunks(m3x_ctx *ctx, const uint8_t *chunks,
+                             uint64_t n_chunks, uint32_t depth,
+                             int64_t mix_len, uint8_t out_root[32]) {
+  if (!ctx) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  void *dev = nullptr;
+  uint64_t bytes = n_chunks * 32;
+  if (bytes) {
+    M3X_HIP_CHECK(hipMalloc(&dev, bytes));
+    if (hipMemcpy(dev, chunks, bytes, hipMemcpyHostToDevice) != hipSuccess) {
+      (void)hipFree(dev);
+      return M3X_ERR_HIP;
+    }
+  }
+  int32_t rc =
+      m3x_merkleize_chunks_dev(ctx, dev, n_chunks, depth, mix_len, out_root);
+  if (dev) (void)hipFree(dev);
+  return rc;
+}
+
+int32_t m3x_validator_subtree_root_dev(m3x_ctx *ctx, const void *ssz_dev,
+                                       uint64_t n, uint32_t depth,
+                                       uint8_t out_root[32]) {
+  if (!ctx || (n > (1ull << depth) && depth < 63)) return M3X_ERR_ARG;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  // leaves
+  uint64_t n_pad = n ? n : 1;
+  int rc = ensure_scratch(ctx, &ctx->scratch_a, &ctx->scratch_a_bytes,
+                          n_pad * 32);
+  if (rc != M3X_OK) return rc;
+  uint8_t *root_dev;
+  M3X_HIP_CHECK(hipMalloc(&root_dev, 64));
+  if (n > 0) {
+    uint32_t blocks = (uint32_t)((n + 255) / 256);
+    hipLaunchKernelGGL(k_validator_leaves, dim3(blocks), dim3(256), 0,
+                       ctx->stream, (const uint8_t *)ssz_dev, n,
+                       ctx->scratch_a);
+    int32_t r = reduce_to_root(ctx, ctx->scratch_a, n, depth, root_dev);
+    if (r != M3X_OK) {
+      (void)hipFree(root_dev);
+      return r;
+    }
+  } else {
+    hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream,
+                       ctx->zeros_dev + 32 * depth, depth, depth, (int64_t)-1,
+                       ctx->zeros_dev, root_dev);
+  }
+  int32_t r2 = M3X_OK;
+  if (hipMemcpyAsync(out_root, root_dev, 32, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    r2 = M3X_ERR_HIP;
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
+  (void)hipFree(root_dev);
+  return r2;
+}
+
+int32_t m3x_merkleize_validators_dev(m3x_ctx *ctx, const void *ssz_dev,
+                                     uint64_t n, uint8_t out_root[32]) {
+  // List[Validator, 2^40]: depth-40 tree + mix_in_length(n)
+  uint8_t sub[32];
+  int32_t rc = m3x_validator_subtree_root_dev(ctx, ssz_dev, n, 40, sub);
+  if (rc != M3X_OK) return rc;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  uint8_t *tmp;
+  M3X_HIP_CHECK(hipMalloc(&tmp, 64));
+  if (hipMemcpy(tmp, sub, 32, hipMemcpyHostToDevice) != hipSuccess) {
+    (void)hipFree(tmp);
+    return M3X_ERR_HIP;
+  }
+  hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream, tmp, 40,
+                     40, (int64_t)n, ctx->zeros_dev, tmp + 32);
+  int32_t r2 = M3X_OK;
+  if (hipMemcpyAsync(out_root, tmp + 32, 32, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    r2 = M3X_ERR_HIP;
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
+  (void)hipFree(tmp);
+  return r2;
+}
+
+int32_t m3x_merkleize_validators(m3x_ctx *ctx, const uint8_t *ssz, uint64_t n,
+                                 uint8_t out_root[32]) {
+  if (!ctx) return M3X_ERR_ARG;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  void *dev = nullptr;
+  uint64_t bytes = n * 121;
+  // pad allocation to a multiple of 4 for the staging loads
+  if (n) {
+    M3X_HIP_CHECK(hipMalloc(&dev, (bytes + 3) & ~3ull));
+    if (hipMemcpy(dev, ssz, bytes, hipMemcpyHostToDevice) != hipSuccess) {
+      (void)hipFree(dev);
+      return M3X_ERR_HIP;
+    }
+  }
+  int32_t rc = m3x_merkleize_validators_dev(ctx, dev, n, out_root);
+  if (dev) (void)hipFree(dev);
+  return rc;
+}
+
+} // extern "C"

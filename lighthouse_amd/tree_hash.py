@@ -1,0 +1,65 @@
+"""Host mirror of the reference's hash_tree_root seam (hot path #2).
+
+Mirrors BeaconState::update_tree_hash_cache /
+update_validators_tree_hash_cache
+(/root/reference/consensus/types/src/beacon_state.rs:2031-2046) and the
+tree_hash packing rules over the m3x C-ABI. GPU-only — no CPU fallback."""
+import ctypes
+
+from . import _native
+
+
+def _ceil_log2(x: int) -> int:
+    d = 0
+    while (1 << d) < x:
+        d += 1
+    return d
+
+
+def validator_registry_root(ssz: bytes, n: int, ctx=None) -> bytes:
+    """hash_tree_root of List[Validator, 2^40] (eth_spec.rs:404) from packed
+    121-byte SSZ records."""
+    ctx = ctx or _native.default_ctx()
+    out = ctypes.create_string_buffer(32)
+    rc = ctx._lib.m3x_merkleize_validators(ctx.handle, ssz, n, out)
+    if rc != 0:
+        raise RuntimeError(f"m3x_merkleize_validators rc={rc}")
+    return out.raw
+
+
+def merkleize_chunks(chunks: bytes, n_chunks: int, depth: int,
+                     mix_len: int = -1, ctx=None) -> bytes:
+    ctx = ctx or _native.default_ctx()
+    out = ctypes.create_string_buffer(32)
+    rc = ctx._lib.m3x_merkleize_chunks(ctx.handle, chunks, n_chunks, depth,
+                                       mix_len, out)
+    if rc != 0:
+        raise RuntimeError(f"m3x_merkleize_chunks rc={rc}")
+    return out.raw
+
+
+def basic_list_root(data: bytes, n_elems: int, elem_size: int,
+                    limit_elems: int, ctx=None) -> bytes:
+    """List of basic elements (u64 balances, u8 participation, ...):
+    LE-packed 32B chunks, depth from the limit, mix_in_length(n)."""
+    limit_chunks = max((limit_elems * elem_size + 31) // 32, 1)
+    n_chunks = (len(data) + 31) // 32
+    padded = data + b"\x00" * (n_chunks * 32 - len(data))
+    return merkleize_chunks(padded, n_chunks, _ceil_log2(limit_chunks),
+                            n_elems, ctx=ctx)
+
+
+def basic_vector_root(data: bytes, n_elems: int, elem_size: int, ctx=None) -> bytes:
+    limit_chunks = max((n_elems * elem_size + 31) // 32, 1)
+    n_chunks = (len(data) + 31) // 32
+    padded = data + b"\x00" * (n_chunks * 32 - len(data))
+    return merkleize_chunks(padded, n_chunks, _ceil_log2(limit_chunks), -1,
+                            ctx=ctx)
+
+
+def root_vector_root(roots: bytes, n: int, ctx=None) -> bytes:
+    return merkleize_chunks(roots, n, _ceil_log2(max(n, 1)), -1, ctx=ctx)
+
+
+def root_list_root(roots: bytes, n: int, limit: int, ctx=None) -> bytes:
+    return merkleize_chunks(roots, n, _ceil_log2(max(limit, 1)), n, ctx=ctx)

@@ -997,5 +997,8 @@ if __name__ == "__main__":
     repo = os.path.dirname(os.path.dirname(here))
     validate()
     emit_consts(os.path.join(repo, "oracle", "bls_consts.h"))
+    # same constants for the HIP product code (kept in csrc so the product
+    # never includes from oracle/)
+    emit_consts(os.path.join(repo, "lighthouse_amd", "csrc", "bls_consts.h"))
     emit_fixtures(os.path.join(here, "bls_fixtures.json"))
     print("all good")
