@@ -1,0 +1,316 @@
+// MI355X-native batched BLS12-381 signature-set verification (hot path #1).
+//
+// Implements the blst.rs:37-119 contract (see include/m3x_consensus.h and
+// DESIGN.md): per set — decompress + subgroup-check sigma (deferred checks
+// exactly as generic_aggregate_signature.rs:161-176 + blst.rs:73-77),
+// aggregate the set's pubkeys, scale by the host-drawn 64-bit r_i, hash the
+// message to G2 (RFC 9380), accumulate one Miller loop per set; then one
+// extra pair e(-g1, sum r_i sigma_i), one final exponentiation, compare to
+// one. CDNA4 shape: one set per lane, kernels split by divergence class,
+// grids ≫256 workgroups; integer VALU workload (no MFMA).
+#include "bls_device.hh"
+#include "m3x_ctx.hh"
+#include "../../include/m3x_consensus.h"
+
+using namespace m3xb;
+
+namespace {
+
+struct BlsWork {
+  g1a *p_scaled;   // [n] r_i * aggregate pubkey (affine)
+  g2a *h2c;        // [n] hash_to_curve(msg)
+  g2j *rsig;       // [n] r_i * sigma (jacobian)
+  fp12m *fparts;   // [n] per-set miller values
+  int *fail;       // [1]
+  g2j *sig_sum;    // [1 + 256 scratch]
+  fp12m *gt_parts; // [256]
+  int *verdict;    // [1]
+};
+
+__device__ void order_be_bytes(uint8_t be[32]) {
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    uint64_t limb = BLS_ORDER[3 - i];
+#pragma unroll
+    for (int j = 0; j < 8; j++) be[8 * i + j] = (uint8_t)(limb >> (56 - 8 * j));
+  }
+}
+
+// ---------------------------------------------------------------- kernels
+
+__global__ void k_bls_pk_decompress(const uint8_t *__restrict__ comp,
+                                    uint64_t n, uint8_t *__restrict__ uncomp,
+                                    int32_t *__restrict__ status) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  g1a p;
+  if (g1_decompress(p, comp + 48 * i) != 0) {
+    status[i] = -1;
+    return;
+  }
+  if (p.inf) { // infinity pubkey rejected (generic_public_key.rs:86-94)
+    status[i] = -2;
+    return;
+  }
+  uint8_t be[32];
+  order_be_bytes(be);
+  g1j t;
+  g1j_mul_be(t, p, be, 32);
+  if (!g1j_is_inf(t)) {
+    status[i] = -4; // subgroup check failed
+    return;
+  }
+  g1_to_uncomp(p, uncomp + 96 * i);
+  status[i] = 0;
+}
+
+// per-set prepare: sig decompress + subgroup check, pk aggregation,
+// r-scaling of both sides
+__global__ void k_bls_prepare(const uint8_t *__restrict__ sigs,
+                              const uint8_t *__restrict__ pks,
+                              const uint32_t *__restrict__ offs,
+                              const uint64_t *__restrict__ rands, uint64_t n,
+                              BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  g2a sig;
+  if (g2_decompress(sig, sigs + 96 * i) != 0 || !g2_in_subgroup(sig)) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  uint32_t k0 = offs[i], k1 = offs[i + 1];
+  if (k1 <= k0) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  g1j apk;
+  fp_zero(apk.x);
+  fp_zero(apk.y);
+  fp_zero(apk.z);
+  for (uint32_t k = k0; k < k1; k++) {
+    g1a pk;
+    if (g1_from_uncomp(pk, pks + 96 * (uint64_t)k) != 0) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    g1j_add_aff(apk, apk, pk);
+  }
+  if (g1j_is_inf(apk)) { // aggregate at infinity -> invalid
+    atomicOr(w.fail, 1);
+    return;
+  }
+  uint8_t rbe[8];
+#pragma unroll
+  for (int b = 0; b < 8; b++) rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
+  g1a apk_a;
+  g1j_to_aff(apk_a, apk);
+  g1j rp;
+  g1j_mul_be(rp, apk_a, rbe, 8);
+  g1j_to_aff(w.p_scaled[i], rp);
+  if (sig.inf) {
+    fp2_zero(w.rsig[i].x);
+    fp2_zero(w.rsig[i].y);
+    fp2_zero(w.rsig[i].z);
+  } else {
+    g2j_mul_be(w.rsig[i], sig, rbe, 8);
+  }
+}
+
+__global__ void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
+                          BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  m3xb::h2c_g2(w.h2c[i], msgs + 32 * i);
+}
+
+__global__ void k_bls_miller(uint64_t n, BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  fp12m f;
+  f12_one(f);
+  if (*w.fail == 0) miller_acc(f, w.p_scaled[i], w.h2c[i]);
+  w.fparts[i] = f;
+}
+
+// single 256-thread block: product of all per-set miller values via LDS tree
+__global__ __launch_bounds__(256) void k_bls_reduce_gt(uint64_t n, BlsWork w) {
+  __shared__ fp12m lds[256];
+  fp12m local;
+  f12_one(local);
+  for (uint64_t i = threadIdx.x; i < n; i += 256) f12_mul(local, local, w.fparts[i]);
+  lds[threadIdx.x] = local;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if ((int)threadIdx.x < s) {
+      fp12m t;
+      f12_mul(t, lds[threadIdx.x], lds[threadIdx.x + s]);
+      lds[threadIdx.x] = t;
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) w.gt_parts[0] = lds[0];
+}
+
+// single 256-thread block: sum of all r_i*sigma_i via LDS tree
+__global__ __launch_bounds__(256) void k_bls_reduce_sig(uint64_t n, BlsWork w) {
+  __shared__ g2j lds[256];
+  g2j local;
+  fp2_zero(local.x);
+  fp2_zero(local.y);
+  fp2_zero(local.z);
+  for (uint64_t i = threadIdx.x; i < n; i += 256) g2j_add(local, local, w.rsig[i]);
+  lds[threadIdx.x] = local;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if ((int)threadIdx.x < s) {
+      g2j t;
+      g2j_add(t, lds[threadIdx.x], lds[threadIdx.x + s]);
+      lds[threadIdx.x] = t;
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) w.sig_sum[0] = lds[0];
+}
+
+// final: f_total *= miller(-g1, sig_sum); final_exp; compare to one
+__global__ void k_bls_finish(BlsWork w) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  if (*w.fail) {
+    *w.verdict = 0;
+    return;
+  }
+  fp12m f = w.gt_parts[0];
+  g2a sig_a;
+  g2j_to_aff(sig_a, w.sig_sum[0]);
+  g1a ng1;
+  g1_gen(ng1);
+  fp_neg(ng1.y, ng1.y);
+  miller_acc(f, ng1, sig_a);
+  fp12m e;
+  final_exp(e, f);
+  *w.verdict = f12_is_one(e) ? 1 : 0;
+}
+
+int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
+               const void *pks_dev, const void *offs_dev,
+               const void *rands_dev, uint64_t n, int32_t *out) {
+  BlsWork w;
+  uint64_t bytes = 0;
+  auto align = [](uint64_t x) { return (x + 255) & ~255ull; };
+  uint64_t off_p = bytes; bytes += align(n * sizeof(g1a));
+  uint64_t off_h = bytes; bytes += align(n * sizeof(g2a));
+  uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
+  uint64_t off_f = bytes; bytes += align(n * sizeof(fp12m));
+  uint64_t off_fail = bytes; bytes += 256;
+  uint64_t off_sum = bytes; bytes += align(sizeof(g2j));
+  uint64_t off_gt = bytes; bytes += align(sizeof(fp12m));
+  uint64_t off_v = bytes; bytes += 256;
+  int rc = m3x::ensure_scratch(ctx, &ctx->scratch_a, &ctx->scratch_a_bytes,
+                               bytes);
+  if (rc != M3X_OK) return rc;
+  uint8_t *base = ctx->scratch_a;
+  w.p_scaled = reinterpret_cast<g1a *>(base + off_p);
+  w.h2c = reinterpret_cast<g2a *>(base + off_h);
+  w.rsig = reinterpret_cast<g2j *>(base + off_r);
+  w.fparts = reinterpret_cast<fp12m *>(base + off_f);
+  w.fail = reinterpret_cast<int *>(base + off_fail);
+  w.sig_sum = reinterpret_cast<g2j *>(base + off_sum);
+  w.gt_parts = reinterpret_cast<fp12m *>(base + off_gt);
+  w.verdict = reinterpret_cast<int *>(base + off_v);
+  M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
+  uint32_t blocks = (uint32_t)((n + 63) / 64);
+  hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
+                     (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
+                     (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
+                     n, w);
+  hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream,
+                     (const uint8_t *)msgs_dev, n, w);
+  hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
+                     w);
+  hipLaunchKernelGGL(k_bls_reduce_gt, dim3(1), dim3(256), 0, ctx->stream, n,
+                     w);
+  hipLaunchKernelGGL(k_bls_reduce_sig, dim3(1), dim3(256), 0, ctx->stream, n,
+                     w);
+  hipLaunchKernelGGL(k_bls_finish, dim3(1), dim3(64), 0, ctx->stream, w);
+  int32_t verdict = 0;
+  M3X_HIP_CHECK(hipMemcpyAsync(&verdict, w.verdict, 4, hipMemcpyDeviceToHost,
+                               ctx->stream));
+  M3X_HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  *out = verdict;
+  return M3X_OK;
+}
+
+} // namespace
+
+extern "C" {
+
+int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp, uint64_t n,
+                              uint8_t *uncomp, int32_t *status) {
+  if (!ctx || n == 0) return M3X_ERR_ARG;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *comp_d, *unc_d;
+  int32_t *st_d;
+  M3X_HIP_CHECK(hipMalloc(&comp_d, n * 48));
+  M3X_HIP_CHECK(hipMalloc(&unc_d, n * 96));
+  M3X_HIP_CHECK(hipMalloc(&st_d, n * 4));
+  hipMemcpyAsync(comp_d, comp, n * 48, hipMemcpyHostToDevice, ctx->stream);
+  uint32_t blocks = (uint32_t)((n + 63) / 64);
+  hipLaunchKernelGGL(k_bls_pk_decompress, dim3(blocks), dim3(64), 0,
+                     ctx->stream, comp_d, n, unc_d, st_d);
+  hipMemcpyAsync(uncomp, unc_d, n * 96, hipMemcpyDeviceToHost, ctx->stream);
+  hipMemcpyAsync(status, st_d, n * 4, hipMemcpyDeviceToHost, ctx->stream);
+  hipError_t e = hipStreamSynchronize(ctx->stream);
+  (void)hipFree(comp_d);
+  (void)hipFree(unc_d);
+  (void)hipFree(st_d);
+  return e == hipSuccess ? M3X_OK : M3X_ERR_HIP;
+}
+
+int32_t m3x_bls_verify_sets_dev(m3x_ctx *ctx, const void *msgs_dev,
+                                const void *sigs_dev, const void *pks_dev,
+                                const void *pk_offsets_dev,
+                                const void *rands_dev, uint64_t n) {
+  if (!ctx) return M3X_ERR_ARG;
+  if (n == 0) return 0; // blst.rs:42-44 (host normally short-circuits)
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  int32_t verdict = 0;
+  int rc = run_verify(ctx, msgs_dev, sigs_dev, pks_dev, pk_offsets_dev,
+                      rands_dev, n, &verdict);
+  if (rc != M3X_OK) return rc;
+  return verdict;
+}
+
+int32_t m3x_bls_verify_sets(m3x_ctx *ctx, const uint8_t *msgs,
+                            const uint8_t *sigs, const uint8_t *pks,
+                            const uint32_t *pk_offsets, const uint64_t *rands,
+                            uint64_t n) {
+  if (!ctx) return M3X_ERR_ARG;
+  if (n == 0) return 0;
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint64_t nk = pk_offsets[n];
+  uint8_t *msgs_d, *sigs_d, *pks_d;
+  uint32_t *offs_d;
+  uint64_t *rands_d;
+  M3X_HIP_CHECK(hipMalloc(&msgs_d, n * 32));
+  M3X_HIP_CHECK(hipMalloc(&sigs_d, n * 96));
+  M3X_HIP_CHECK(hipMalloc(&pks_d, nk * 96));
+  M3X_HIP_CHECK(hipMalloc(&offs_d, (n + 1) * 4));
+  M3X_HIP_CHECK(hipMalloc(&rands_d, n * 8));
+  hipMemcpy(msgs_d, msgs, n * 32, hipMemcpyHostToDevice);
+  hipMemcpy(sigs_d, sigs, n * 96, hipMemcpyHostToDevice);
+  hipMemcpy(pks_d, pks, nk * 96, hipMemcpyHostToDevice);
+  hipMemcpy(offs_d, pk_offsets, (n + 1) * 4, hipMemcpyHostToDevice);
+  hipMemcpy(rands_d, rands, n * 8, hipMemcpyHostToDevice);
+  int32_t rc = m3x_bls_verify_sets_dev(ctx, msgs_d, sigs_d, pks_d, offs_d,
+                                       rands_d, n);
+  (void)hipFree(msgs_d);
+  (void)hipFree(sigs_d);
+  (void)hipFree(pks_d);
+  (void)hipFree(offs_d);
+  (void)hipFree(rands_d);
+  return rc;
+}
+
+} // extern "C"

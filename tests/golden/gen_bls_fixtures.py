@@ -807,7 +807,7 @@ def limbs(v, n=6):
 
 def c_limbs(name, v, n=6):
     ls = ", ".join(f"0x{w:016x}ULL" for w in limbs(v, n))
-    return f"static const uint64_t {name}[{n}] = {{{ls}}};\n"
+    return f"static constexpr uint64_t {name}[{n}] = {{{ls}}};\n"
 
 
 def c_fp2(name, v):
@@ -830,7 +830,7 @@ def emit_consts(path):
     out.append("#include <stdint.h>\n")
     out.append(c_limbs("BLS_P", P))
     out.append(c_limbs("BLS_R2", R2))  # Montgomery R^2 mod p
-    out.append(f"static const uint64_t BLS_N0 = 0x{N0:016x}ULL;\n")
+    out.append(f"static constexpr uint64_t BLS_N0 = 0x{N0:016x}ULL;\n")
     out.append(c_limbs("BLS_ORDER", R, 4))
     out.append(f"#define BLS_X_ABS 0x{H_EFF_ABS:016x}ULL\n")
     out.append("#define BLS_X_NEGATIVE 1\n")

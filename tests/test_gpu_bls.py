@@ -1,0 +1,169 @@
+"""GPU parity tests for the BLS batch-verify hot path: HIP kernels vs the
+golden fixtures (Python reference pinned to the reference repo's interop
+vectors) and vs the C oracle, including the rejection rules of
+blst.rs:37-119 and the behavioral cases of crypto/bls/tests/tests.rs."""
+import ctypes
+import hashlib
+import json
+from pathlib import Path
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+FIXTURES = json.loads(
+    (Path(__file__).parent / "golden" / "bls_fixtures.json").read_text()
+)
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    from lighthouse_amd import _native
+
+    return _native.default_ctx()
+
+
+def _mk_sets(case):
+    from lighthouse_amd import bls
+
+    return [
+        bls.SignatureSet(
+            signature=bls.Signature.from_compressed(
+                bytes.fromhex(s["sig_compressed_hex"])
+            ),
+            signing_keys=[
+                bls.PublicKey.from_uncompressed(bytes.fromhex(p))
+                for p in s["pks_uncompressed_hex"]
+            ],
+            message=bytes.fromhex(s["msg_hex"]),
+        )
+        for s in case["sets"]
+    ]
+
+
+def test_pk_decompress_parity(ctx):
+    from lighthouse_amd import bls
+
+    comp = b"".join(
+        bytes.fromhex(it["pk_compressed_hex"]) for it in FIXTURES["interop"]
+    )
+    out, status = bls.decompress_pubkeys(comp, 10, ctx=ctx)
+    for i, it in enumerate(FIXTURES["interop"]):
+        assert status[i] == 0
+        assert out[96 * i : 96 * (i + 1)].hex() == it["pk_uncompressed_hex"]
+    # infinity and garbage rejected
+    bad = bytes([0xC0] + [0] * 47) + b"\x80" + b"\xff" * 47
+    _, st = bls.decompress_pubkeys(bad, 2, ctx=ctx)
+    assert st[0] != 0 and st[1] != 0
+
+
+def test_batch_cases_golden(ctx):
+    from lighthouse_amd import bls
+
+    for case in FIXTURES["batch_cases"]:
+        sets = _mk_sets(case)
+        rands = [int(r) for r in case["rands"]]
+        got = bls.verify_signature_sets(sets, ctx=ctx, _rands=rands)
+        assert got == bool(case["verdict"]), case["name"]
+
+
+def test_batch_random_rands_still_correct(ctx):
+    # verdicts must be rand-independent (with overwhelming probability)
+    from lighthouse_amd import bls
+
+    for case in FIXTURES["batch_cases"]:
+        got = bls.verify_signature_sets(_mk_sets(case), ctx=ctx)
+        assert got == bool(case["verdict"]), case["name"]
+
+
+def test_host_rules(ctx):
+    from lighthouse_amd import bls
+
+    # empty list -> False (blst.rs:42-44)
+    assert bls.verify_signature_sets([], ctx=ctx) is False
+    # empty signature -> False (blst.rs:80-83)
+    pk = bls.PublicKey.from_uncompressed(
+        bytes.fromhex(FIXTURES["interop"][0]["pk_uncompressed_hex"])
+    )
+    s = bls.SignatureSet(bls.Signature.empty(), [pk], b"\x00" * 32)
+    assert bls.verify_signature_sets([s], ctx=ctx) is False
+    # empty signing keys -> False (blst.rs:86-89)
+    sig = bls.Signature.from_compressed(
+        bytes.fromhex(FIXTURES["signatures"][0]["sig_compressed_hex"])
+    )
+    s = bls.SignatureSet(sig, [], b"\x00" * 32)
+    assert bls.verify_signature_sets([s], ctx=ctx) is False
+    # malformed signature bytes -> False (not an exception)
+    bad = bytearray(bytes.fromhex(FIXTURES["signatures"][0]["sig_compressed_hex"]))
+    bad[5] ^= 0xFF
+    s = bls.SignatureSet(
+        bls.Signature.from_compressed(bytes(bad)), [pk], b"\x00" * 32
+    )
+    # either decompress fails or point is wrong: must be False either way
+    assert bls.verify_signature_sets([s], ctx=ctx) is False
+
+
+def test_batch_vs_oracle_random_sets(ctx, oracle):
+    """64 synthetic k=1 sets signed with interop keys: GPU verdict must agree
+    with the C oracle on the same inputs and rands."""
+    from lighthouse_amd import bls
+
+    n = 64
+    msgs = b""
+    sigs = b""
+    pks = b""
+    offs = [0]
+    sk = ctypes.create_string_buffer(32)
+    sig = ctypes.create_string_buffer(96)
+    sets = []
+    for i in range(n):
+        idx = i % 10
+        msg = hashlib.sha256(b"gpu%d" % i).digest()
+        oracle.m3x_oracle_bls_keygen(ctypes.c_uint64(idx), sk)
+        assert oracle.m3x_oracle_bls_sign(sk.raw, msg, sig) == 0
+        msgs += msg
+        sigs += sig.raw
+        pk_unc = bytes.fromhex(FIXTURES["interop"][idx]["pk_uncompressed_hex"])
+        pks += pk_unc
+        offs.append(offs[-1] + 1)
+        sets.append(
+            bls.SignatureSet(
+                bls.Signature.from_compressed(sig.raw),
+                [bls.PublicKey.from_uncompressed(pk_unc)],
+                msg,
+            )
+        )
+    rands = [(i * 2654435761 + 7) | 1 for i in range(n)]
+    got = bls.verify_signature_sets(sets, ctx=ctx, _rands=rands)
+    off_arr = (ctypes.c_uint32 * (n + 1))(*offs)
+    rand_arr = (ctypes.c_uint64 * n)(*rands)
+    want = oracle.m3x_oracle_bls_verify_sets(
+        msgs, sigs, pks, off_arr, rand_arr, ctypes.c_uint64(n)
+    )
+    assert got is True and want == 1
+
+    # flip one message: both must reject
+    bad_msgs = b"\xaa" * 32 + msgs[32:]
+    sets[0] = bls.SignatureSet(sets[0].signature, sets[0].signing_keys, b"\xaa" * 32)
+    got = bls.verify_signature_sets(sets, ctx=ctx, _rands=rands)
+    want = oracle.m3x_oracle_bls_verify_sets(
+        bad_msgs, sigs, pks, off_arr, rand_arr, ctypes.c_uint64(n)
+    )
+    assert got is False and want == 0
+
+
+def test_aggregate_sets_k_gt_1(ctx, oracle):
+    """aggregate sets (k up to 10) — gossip-aggregate shape."""
+    from lighthouse_amd import bls
+
+    case = next(
+        c for c in FIXTURES["batch_cases"] if c["name"] == "three_valid_mixed_k"
+    )
+    sets = _mk_sets(case)
+    assert bls.verify_signature_sets(sets, ctx=ctx) is True
+    # swap a signer's key for another: must fail
+    other = bls.PublicKey.from_uncompressed(
+        bytes.fromhex(FIXTURES["interop"][9]["pk_uncompressed_hex"])
+    )
+    sets[1].signing_keys[0] = other
+    assert bls.verify_signature_sets(sets, ctx=ctx) is False
