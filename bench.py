@@ -1,0 +1,367 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json workloads on MI355X.
+
+A "step" is one pass of the hot path over one batch of synthetic input:
+  - primary (BASELINE configs[1], the quoted single-GPU workload): batch
+    BLS verification of 65,536 attestation signature sets (75% k=1
+    unaggregated + 25% aggregates with k=512 — committee size at 2^20
+    active validators), inputs resident in HBM when the timed region
+    starts; each step runs the full blst.rs:37-119 check incl. the final
+    exponentiation and verdict readback.
+  - in the same step, the SHA256 path (configs[2]): full rebuild of the
+    1,048,576-validator registry hash_tree_root (9,437,204 two-to-one node
+    hashes), reported as extra_metrics.
+Multi-GPU (--gpus N via torch.distributed.run): weak scaling — each rank
+verifies its own 64k-set batch (verdicts AND-combined with one 4-byte
+all_reduce(MIN) over RCCL) and computes its aligned shard of ONE shared
+1M-validator registry (subtree roots gathered to rank 0 which finishes the
+zero cap + mix_in_length).
+
+Workload generation (untimed) uses the CPU oracle as the synthetic-data
+SIGNER (the r-side of each set must verify, so sets are signed with
+aggregate secrets derived from the interop keypool); the timed region runs
+only the product HIP path. The cpu_baseline leg times the same oracle as
+the reference-equivalent CPU port (kind "port", OpenMP, cores stated).
+"""
+import argparse
+import ctypes
+import hashlib
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO))
+
+# BLS12-381 group order (for synthetic aggregate secret keys)
+ORDER = 0x73EDA753299D7D483339D80809A1D80553BDA402FFFE5BFEFFFFFFFF00000001
+
+# analytic roofline constants (derivation in DESIGN.md §Roofline):
+# Fp multiplies per set in the dominant kernel (k_bls_miller): 63 doubling
+# steps x (f^2 108 + sparse line mult 54 + line coeffs 36 + point dbl 21)
+# + 5 addition steps x ~110  ~= 14.3k; int32 ALU ops per 6x64-limb CIOS
+# Montgomery multiply ~= 550 (144 32-bit mults + carries/adds).
+FP_MUL_PER_MILLER = 14300
+INT_OPS_PER_FP_MUL = 550
+PEAK_INT32_OPS = 256 * 4 * 32 * 2.4e9  # CUs x SIMDs x lanes x clock
+SHA_OPS_PER_NODE = 1900  # two compressions, pad block constant-folded
+REGISTRY_NODE_HASHES = 8 * (1 << 20) + ((1 << 20) - 1) + 20 + 1  # 9,437,204
+
+POOL = 4096
+N_SETS = 65536
+N_AGG = N_SETS // 4
+K_AGG = 512
+N_VALIDATORS = 1 << 20
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def build_bls_workload(oracle, seed=0xC0FFEE):
+    """65,536 sets: 49,152 k=1 + 16,384 k=512 aggregates over a 4096-key
+    interop pool; every set carries a real signature (signed with the sum
+    of the member secret keys), so the batch verdict is True."""
+    t0 = time.time()
+    sks = ctypes.create_string_buffer(32 * POOL)
+    pks = ctypes.create_string_buffer(96 * POOL)
+    oracle.m3x_oracle_bls_keypool(ctypes.c_uint64(POOL), sks, pks)
+    sk_ints = [
+        int.from_bytes(sks.raw[32 * i : 32 * (i + 1)], "big") for i in range(POOL)
+    ]
+    msgs = bytearray()
+    sign_sks = bytearray()
+    pk_bytes = bytearray()
+    offsets = [0]
+    costs = []
+    for i in range(N_SETS):
+        msg = hashlib.sha256(b"c2msg%d" % i).digest()
+        msgs += msg
+        if i < N_SETS - N_AGG:  # k = 1
+            j = i % POOL
+            sign_sks += sk_ints[j].to_bytes(32, "big")
+            pk_bytes += pks.raw[96 * j : 96 * (j + 1)]
+            offsets.append(offsets[-1] + 1)
+            costs.append(1)
+        else:  # aggregate, k = 512, contiguous committee window
+            off = (i * 37) % (POOL - K_AGG)
+            agg = 0
+            for j in range(off, off + K_AGG):
+                agg = (agg + sk_ints[j]) % ORDER
+            sign_sks += agg.to_bytes(32, "big")
+            pk_bytes += pks.raw[96 * off : 96 * (off + K_AGG)]
+            offsets.append(offsets[-1] + K_AGG)
+            costs.append(K_AGG)
+    sigs = ctypes.create_string_buffer(96 * N_SETS)
+    rc = oracle.m3x_oracle_bls_sign_batch(
+        ctypes.c_uint64(N_SETS), bytes(sign_sks), bytes(msgs), sigs
+    )
+    assert rc == 0
+    rands = [((i * 0x9E3779B97F4A7C15 + seed) | 1) & 0xFFFFFFFFFFFFFFFF for i in range(N_SETS)]
+    log(f"bls workload built in {time.time()-t0:.1f}s")
+    return {
+        "msgs": bytes(msgs),
+        "sigs": sigs.raw,
+        "pks": bytes(pk_bytes),
+        "offsets": offsets,
+        "rands": rands,
+        "costs": costs,
+    }
+
+
+def upload_bls(ctx, w):
+    import numpy as np
+
+    dev = {}
+    dev["msgs"] = ctx.upload(w["msgs"])
+    dev["sigs"] = ctx.upload(w["sigs"])
+    dev["pks"] = ctx.upload(w["pks"])
+    dev["offsets"] = ctx.upload(np.asarray(w["offsets"], dtype=np.uint32).tobytes())
+    dev["rands"] = ctx.upload(np.asarray(w["rands"], dtype=np.uint64).tobytes())
+    return dev
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+        tdist.init_process_group("nccl")
+
+    os.environ.setdefault("M3X_DEVICE", str(local_rank))
+    from lighthouse_amd import _native, state
+
+    ctx = _native.Ctx(local_rank)
+    oracle = ctypes.CDLL(str(REPO / "oracle" / "liboracle.so"))
+
+    # ---------------- workload prep (untimed) ----------------
+    bls_w = build_bls_workload(oracle, seed=0xC0FFEE + rank)
+    bls_dev = upload_bls(ctx, bls_w)
+
+    t0 = time.time()
+    ssz_full = state.validators_ssz(N_VALIDATORS)
+    log(f"state generated in {time.time()-t0:.1f}s")
+    # merkle shard for this rank (whole registry at N=1)
+    start, per = rank * (N_VALIDATORS // world), N_VALIDATORS // world
+    sub_depth = per.bit_length() - 1
+    ssz_shard = ssz_full[121 * start : 121 * (start + per)]
+    ssz_dev = ctx.upload(ssz_shard + b"\x00" * 4)
+
+    lib = ctx._lib
+    out32 = ctypes.create_string_buffer(32)
+
+    def bls_step():
+        v = lib.m3x_bls_verify_sets_dev(
+            ctx.handle,
+            bls_dev["msgs"],
+            bls_dev["sigs"],
+            bls_dev["pks"],
+            bls_dev["offsets"],
+            bls_dev["rands"],
+            N_SETS,
+        )
+        if world > 1:
+            t = torch.tensor([v], dtype=torch.int32, device="cuda")
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            v = int(t.item())
+        return v
+
+    def merkle_step():
+        rc = lib.m3x_validator_subtree_root_dev(
+            ctx.handle, ssz_dev, per, sub_depth, out32
+        )
+        assert rc == 0, rc
+        node = out32.raw
+        if world > 1:
+            t = torch.frombuffer(bytearray(node), dtype=torch.uint8).clone().cuda()
+            gathered = (
+                [torch.zeros_like(t) for _ in range(world)] if rank == 0 else None
+            )
+            dist.gather(t, gathered, dst=0)
+            if rank == 0:
+                nodes = [bytes(g.cpu().numpy().tobytes()) for g in gathered]
+                level = sub_depth
+                while len(nodes) > 1:
+                    nxt = []
+                    for i in range(0, len(nodes), 2):
+                        o = ctypes.create_string_buffer(32)
+                        lib.m3x_merkleize_chunks(
+                            ctx.handle, nodes[i] + nodes[i + 1], 2, 1, -1, o
+                        )
+                        nxt.append(o.raw)
+                    nodes = nxt
+                    level += 1
+                return ctx.finalize_root(nodes[0], level, 40, N_VALIDATORS)
+            return None
+        return ctx.finalize_root(node, sub_depth, 40, N_VALIDATORS)
+
+    # correctness gate before timing: verdict true, root matches oracle
+    assert bls_step() == 1, "bls batch verdict false on valid workload"
+    root = merkle_step()
+    if rank == 0:
+        want = ctypes.create_string_buffer(32)
+        oracle.m3x_oracle_validator_registry_root(
+            ssz_full, ctypes.c_uint64(N_VALIDATORS), want
+        )
+        assert root == want.raw, "registry root mismatch vs oracle"
+        log("correctness gate passed (verdict true, root bit-exact)")
+
+    # ---------------- timed region ----------------
+    for _ in range(args.warmup):
+        bls_step()
+        merkle_step()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    ctx.timing_enable(True)
+    t0 = time.time()
+    for _ in range(args.steps):
+        v = bls_step()
+        assert v == 1
+        merkle_step()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.time() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    ktimes = ctx.kernel_times()
+
+    # split timing: one more pass of each, timed separately (for extras)
+    torch.cuda.synchronize()
+    tb = time.time()
+    bls_step()
+    torch.cuda.synchronize()
+    bls_only = time.time() - tb
+    tb = time.time()
+    merkle_step()
+    torch.cuda.synchronize()
+    merkle_only = time.time() - tb
+
+    # ---------------- cpu baseline (rank 0, N=1) ----------------
+    cpu_baseline = None
+    cpu_sha = None
+    if rank == 0 and world == 1:
+        # BLS: 256-set sample, same 75/25 mix (192 k=1 + 64 k=512)
+        idx = list(range(192)) + list(range(N_SETS - N_AGG, N_SETS - N_AGG + 64))
+        msgs = b"".join(bls_w["msgs"][32 * i : 32 * (i + 1)] for i in idx)
+        sigs = b"".join(bls_w["sigs"][96 * i : 96 * (i + 1)] for i in idx)
+        pkb = b""
+        offs = [0]
+        for i in idx:
+            a, b = bls_w["offsets"][i], bls_w["offsets"][i + 1]
+            pkb += bls_w["pks"][96 * a : 96 * b]
+            offs.append(offs[-1] + (b - a))
+        off_arr = (ctypes.c_uint32 * len(offs))(*offs)
+        rnd = (ctypes.c_uint64 * len(idx))(*[bls_w["rands"][i] for i in idx])
+        tb = time.time()
+        vcpu = oracle.m3x_oracle_bls_verify_sets(
+            msgs, sigs, pkb, off_arr, rnd, ctypes.c_uint64(len(idx))
+        )
+        tcpu = time.time() - tb
+        assert vcpu == 1
+        cores = os.cpu_count()
+        cpu_baseline = {
+            "value": len(idx) / tcpu,
+            "unit": "sets/s",
+            "cores": cores,
+            "kind": "port",
+            "sample": f"{len(idx)} sets (75% k=1, 25% k=512), {tcpu:.1f}s, OpenMP x{cores}",
+        }
+        # SHA: full registry once
+        tb = time.time()
+        want = ctypes.create_string_buffer(32)
+        oracle.m3x_oracle_validator_registry_root(
+            ssz_full, ctypes.c_uint64(N_VALIDATORS), want
+        )
+        tsha = time.time() - tb
+        cpu_sha = {
+            "node_hashes_per_sec": REGISTRY_NODE_HASHES / tsha,
+            "state_root_s": tsha,
+            "cores": cores,
+        }
+
+    # ---------------- report ----------------
+    if rank == 0:
+        steps = args.steps
+        sets_per_step = N_SETS * world  # weak scaling: each rank its batch
+        value = sets_per_step * steps / elapsed
+        miller_ms, miller_n = ktimes.get("bls_miller", (0.0, 0))
+        roofline = None
+        if miller_n:
+            per_launch_ms = miller_ms / miller_n
+            ops = N_SETS * FP_MUL_PER_MILLER * INT_OPS_PER_FP_MUL
+            achieved = ops / (per_launch_ms / 1e3)
+            roofline = {
+                "bound": "valu",
+                "achieved": achieved,
+                "peak": PEAK_INT32_OPS,
+                "unit": "int32 ops/s",
+                "frac": achieved / PEAK_INT32_OPS,
+                "traffic": None,
+                "kernel": "k_bls_miller",
+                "basis": "analytic op count (DESIGN.md Roofline); VALU workload per SURVEY 8d — not HBM/MFMA-bound",
+            }
+        leaves_ms, leaves_n = ktimes.get("leaves", (0.0, 0))
+        reduce_ms, reduce_n = ktimes.get("reduce", (0.0, 0))
+        sha_hps = (
+            REGISTRY_NODE_HASHES / merkle_only if merkle_only > 0 else None
+        )
+        line = {
+            "metric": "bls_sig_sets_verified_per_sec",
+            "value": value,
+            "unit": "sets/s",
+            "n_gpus": n_gpus,
+            "steps": steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u64",
+            "data": "synthetic",
+            "config": {
+                "workload": "c2_64k_attestation_sets_plus_c3_registry_root",
+                "n_sets_per_gpu": N_SETS,
+                "set_mix": "75% k=1 unagg, 25% aggregates k=512",
+                "n_validators": N_VALIDATORS,
+                "parallelism": f"shard{world}" if world > 1 else "single",
+            },
+            "extra_metrics": {
+                "sha256_node_hashes_per_sec": sha_hps,
+                "state_root_registry_ms": merkle_only * 1e3,
+                "bls_batch_ms": bls_only * 1e3,
+                "kernel_ms_per_step": {
+                    k: v[0] / max(steps, 1) for k, v in ktimes.items() if v[1]
+                },
+                "cpu_sha_baseline": cpu_sha,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

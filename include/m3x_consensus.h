@@ -46,6 +46,21 @@ void m3x_ctx_destroy(m3x_ctx *ctx);
 /* version/capability probe (also serves as a loadability check) */
 int32_t m3x_abi_version(void);
 
+/* per-kernel HIP-event timing (for benchmarks/roofline): enable resets the
+ * accumulators; kernel ids are the M3X_K_* slots (0 leaves, 1 reduce,
+ * 2 finalize, 3 bls_prepare, 4 bls_h2c, 5 bls_miller, 6 bls_reduce,
+ * 7 bls_finish). */
+int32_t m3x_timing_enable(m3x_ctx *ctx, int32_t on);
+int32_t m3x_kernel_ms(m3x_ctx *ctx, int32_t kernel_id, double *ms,
+                      uint64_t *launches);
+
+/* carry a 32B subtree root from `from_level` to `to_depth` against the
+ * zero ladder, then optionally mix_in_length — the multi-GPU cap-finishing
+ * primitive paired with m3x_validator_subtree_root_dev. */
+int32_t m3x_finalize_root(m3x_ctx *ctx, const uint8_t node[32],
+                          uint32_t from_level, uint32_t to_depth,
+                          int64_t mix_len, uint8_t out_root[32]);
+
 /* ---- device buffer management (for *_dev calls / benchmarks) ---- */
 int32_t m3x_dev_alloc(m3x_ctx *ctx, uint64_t bytes, void **dev_ptr);
 int32_t m3x_dev_free(m3x_ctx *ctx, void *dev_ptr);

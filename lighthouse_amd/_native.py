@@ -82,6 +82,27 @@ def _bind(lib):
                 ctypes.c_char_p,
             ],
         ),
+        ("m3x_timing_enable", [ctypes.c_void_p, ctypes.c_int32]),
+        (
+            "m3x_kernel_ms",
+            [
+                ctypes.c_void_p,
+                ctypes.c_int32,
+                ctypes.POINTER(ctypes.c_double),
+                ctypes.POINTER(ctypes.c_uint64),
+            ],
+        ),
+        (
+            "m3x_finalize_root",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_uint32,
+                ctypes.c_uint32,
+                ctypes.c_int64,
+                ctypes.c_char_p,
+            ],
+        ),
         (
             "m3x_bls_pk_decompress",
             [
@@ -192,6 +213,39 @@ class Ctx:
         p = self.alloc(max(len(host_bytes), 4))
         self.h2d(p, host_bytes)
         return p
+
+    # ---- kernel timing (bench/roofline) ----
+    def timing_enable(self, on: bool = True):
+        self._lib.m3x_timing_enable(self._h, 1 if on else 0)
+
+    KERNEL_NAMES = [
+        "leaves",
+        "reduce",
+        "finalize",
+        "bls_prepare",
+        "bls_h2c",
+        "bls_miller",
+        "bls_reduce",
+        "bls_finish",
+    ]
+
+    def kernel_times(self):
+        out = {}
+        for i, name in enumerate(self.KERNEL_NAMES):
+            ms = ctypes.c_double()
+            n = ctypes.c_uint64()
+            if self._lib.m3x_kernel_ms(self._h, i, ctypes.byref(ms), ctypes.byref(n)) == 0:
+                out[name] = (ms.value, n.value)
+        return out
+
+    def finalize_root(self, node: bytes, from_level: int, to_depth: int,
+                      mix_len: int = -1) -> bytes:
+        out = ctypes.create_string_buffer(32)
+        rc = self._lib.m3x_finalize_root(self._h, node, from_level, to_depth,
+                                         mix_len, out)
+        if rc != 0:
+            raise RuntimeError(f"m3x_finalize_root rc={rc}")
+        return out.raw
 
 
 _default_ctx = None

@@ -219,19 +219,29 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.verdict = reinterpret_cast<int *>(base + off_v);
   M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
   uint32_t blocks = (uint32_t)((n + 63) / 64);
+  m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
   hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
                      (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
                      n, w);
+  m3x::time_end(ctx, M3X_K_BLS_PREPARE);
+  m3x::time_begin(ctx, M3X_K_BLS_H2C);
   hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint8_t *)msgs_dev, n, w);
+  m3x::time_end(ctx, M3X_K_BLS_H2C);
+  m3x::time_begin(ctx, M3X_K_BLS_MILLER);
   hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
                      w);
+  m3x::time_end(ctx, M3X_K_BLS_MILLER);
+  m3x::time_begin(ctx, M3X_K_BLS_REDUCE);
   hipLaunchKernelGGL(k_bls_reduce_gt, dim3(1), dim3(256), 0, ctx->stream, n,
                      w);
   hipLaunchKernelGGL(k_bls_reduce_sig, dim3(1), dim3(256), 0, ctx->stream, n,
                      w);
+  m3x::time_end(ctx, M3X_K_BLS_REDUCE);
+  m3x::time_begin(ctx, M3X_K_BLS_FINISH);
   hipLaunchKernelGGL(k_bls_finish, dim3(1), dim3(64), 0, ctx->stream, w);
+  m3x::time_end(ctx, M3X_K_BLS_FINISH);
   int32_t verdict = 0;
   M3X_HIP_CHECK(hipMemcpyAsync(&verdict, w.verdict, 4, hipMemcpyDeviceToHost,
                                ctx->stream));

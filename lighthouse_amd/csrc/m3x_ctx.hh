@@ -4,6 +4,19 @@
 #include <stdint.h>
 #include <mutex>
 
+// named kernel slots for per-kernel HIP-event timing (m3x_kernel_ms)
+enum m3x_kernel_id {
+  M3X_K_LEAVES = 0,
+  M3X_K_REDUCE = 1,
+  M3X_K_FINALIZE = 2,
+  M3X_K_BLS_PREPARE = 3,
+  M3X_K_BLS_H2C = 4,
+  M3X_K_BLS_MILLER = 5,
+  M3X_K_BLS_REDUCE = 6,
+  M3X_K_BLS_FINISH = 7,
+  M3X_K_COUNT = 8
+};
+
 struct m3x_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
@@ -14,8 +27,21 @@ struct m3x_ctx {
   uint64_t scratch_a_bytes = 0;
   uint8_t *scratch_b = nullptr;
   uint64_t scratch_b_bytes = 0;
+  // per-kernel cumulative time (ms) + launch counts since last reset,
+  // measured with hipEvents on `stream`
+  hipEvent_t ev_a[M3X_K_COUNT] = {};
+  hipEvent_t ev_b[M3X_K_COUNT] = {};
+  double kernel_ms[M3X_K_COUNT] = {};
+  uint64_t kernel_launches[M3X_K_COUNT] = {};
+  bool timing = false;
   std::mutex mu; // one context serializes its own calls
 };
+
+namespace m3x {
+// wrap a launch with events when ctx->timing (call after the launch)
+void time_begin(m3x_ctx *ctx, int k);
+void time_end(m3x_ctx *ctx, int k);
+} // namespace m3x
 
 #define M3X_HIP_CHECK(expr)                                                    \
   do {                                                                         \

@@ -208,6 +208,25 @@ __global__ void k_finalize(const uint8_t *__restrict__ node_in,
 // ---------------------------------------------------------------- host side
 
 namespace m3x {
+void time_begin(m3x_ctx *ctx, int k) {
+  if (!ctx->timing) return;
+  if (!ctx->ev_a[k]) {
+    (void)hipEventCreate(&ctx->ev_a[k]);
+    (void)hipEventCreate(&ctx->ev_b[k]);
+  }
+  (void)hipEventRecord(ctx->ev_a[k], ctx->stream);
+}
+
+void time_end(m3x_ctx *ctx, int k) {
+  if (!ctx->timing) return;
+  (void)hipEventRecord(ctx->ev_b[k], ctx->stream);
+  (void)hipEventSynchronize(ctx->ev_b[k]);
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, ctx->ev_a[k], ctx->ev_b[k]);
+  ctx->kernel_ms[k] += ms;
+  ctx->kernel_launches[k]++;
+}
+
 int ensure_scratch(m3x_ctx *ctx, uint8_t **buf, uint64_t *cur,
                    uint64_t bytes) {
   if (*cur >= bytes) return M3X_OK;
@@ -223,6 +242,50 @@ int ensure_scratch(m3x_ctx *ctx, uint8_t **buf, uint64_t *cur,
 extern "C" {
 
 int32_t m3x_abi_version(void) { return 1; }
+
+int32_t m3x_timing_enable(m3x_ctx *ctx, int32_t on) {
+  if (!ctx) return M3X_ERR_ARG;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  ctx->timing = on != 0;
+  for (int i = 0; i < M3X_K_COUNT; i++) {
+    ctx->kernel_ms[i] = 0;
+    ctx->kernel_launches[i] = 0;
+  }
+  return M3X_OK;
+}
+
+int32_t m3x_kernel_ms(m3x_ctx *ctx, int32_t kernel_id, double *ms,
+                      uint64_t *launches) {
+  if (!ctx || kernel_id < 0 || kernel_id >= M3X_K_COUNT) return M3X_ERR_ARG;
+  *ms = ctx->kernel_ms[kernel_id];
+  *launches = ctx->kernel_launches[kernel_id];
+  return M3X_OK;
+}
+
+/* carry a 32B node from `from_level` to `to_depth` against the zero ladder,
+ * then optionally mix_in_length — the multi-GPU cap-finishing primitive. */
+int32_t m3x_finalize_root(m3x_ctx *ctx, const uint8_t node[32],
+                          uint32_t from_level, uint32_t to_depth,
+                          int64_t mix_len, uint8_t out_root[32]) {
+  if (!ctx) return M3X_ERR_ARG;
+  std::lock_guard<std::mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *tmp;
+  M3X_HIP_CHECK(hipMalloc(&tmp, 64));
+  if (hipMemcpy(tmp, node, 32, hipMemcpyHostToDevice) != hipSuccess) {
+    (void)hipFree(tmp);
+    return M3X_ERR_HIP;
+  }
+  hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream, tmp,
+                     from_level, to_depth, mix_len, ctx->zeros_dev, tmp + 32);
+  int32_t r2 = M3X_OK;
+  if (hipMemcpyAsync(out_root, tmp + 32, 32, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    r2 = M3X_ERR_HIP;
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
+  (void)hipFree(tmp);
+  return r2;
+}
 
 int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
   auto *ctx = new m3x_ctx();
@@ -311,9 +374,11 @@ static int32_t reduce_to_root(m3x_ctx *ctx, const uint8_t *nodes_dev,
     uint64_t span = 1ull << levels;
     uint64_t n_out = (m + span - 1) / span;
     uint8_t *dst = (cur == ping) ? pong : ping;
+    m3x::time_begin(ctx, M3X_K_REDUCE);
     hipLaunchKernelGGL(k_reduce, dim3((uint32_t)n_out), dim3(256), 0,
                        ctx->stream, cur, m, dst, levels, level,
                        ctx->zeros_dev);
+    m3x::time_end(ctx, M3X_K_REDUCE);
     cur = dst;
     m = n_out;
     level += levels;
@@ -394,9 +459,11 @@ int32_t m3x_validator_subtree_root_dev(m3x_ctx *ctx, const void *ssz_dev,
   M3X_HIP_CHECK(hipMalloc(&root_dev, 64));
   if (n > 0) {
     uint32_t blocks = (uint32_t)((n + 255) / 256);
+    m3x::time_begin(ctx, M3X_K_LEAVES);
     hipLaunchKernelGGL(k_validator_leaves, dim3(blocks), dim3(256), 0,
                        ctx->stream, (const uint8_t *)ssz_dev, n,
                        ctx->scratch_a);
+    m3x::time_end(ctx, M3X_K_LEAVES);
     int32_t r = reduce_to_root(ctx, ctx->scratch_a, n, depth, root_dev);
     if (r != M3X_OK) {
       (void)hipFree(root_dev);

@@ -219,6 +219,41 @@ int m3x_oracle_bls_verify_sets(const uint8_t *msgs, const uint8_t *sigs,
   return fp12_is_one(&e);
 }
 
+int m3x_oracle_bls_keypool(uint64_t n, uint8_t *sks /* n*32 */,
+                           uint8_t *pks_uncomp /* n*96 */) {
+  /* interop keypairs 0..n-1 (lib.rs:40-55), pubkeys uncompressed —
+   * OpenMP-parallel workload generator for tests/bench. */
+  bls_init();
+  int fail = 0;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 16)
+#endif
+  for (int64_t i = 0; i < (int64_t)n; i++) {
+    m3x_oracle_bls_keygen((uint64_t)i, sks + 32 * i);
+    g1_jac_t j;
+    g1_mul_be(&j, &G1_GEN, sks + 32 * i, 32);
+    g1_aff_t a;
+    g1_to_aff(&a, &j);
+    g1_to_uncomp(&a, pks_uncomp + 96 * i);
+  }
+  return fail;
+}
+
+int m3x_oracle_bls_sign_batch(uint64_t n, const uint8_t *sks /* n*32 */,
+                              const uint8_t *msgs /* n*32 */,
+                              uint8_t *sigs /* n*96 */) {
+  bls_init();
+  int fail = 0;
+#ifdef _OPENMP
+#pragma omp parallel for schedule(dynamic, 4)
+#endif
+  for (int64_t i = 0; i < (int64_t)n; i++) {
+    if (m3x_oracle_bls_sign(sks + 32 * i, msgs + 32 * i, sigs + 96 * i) != 0)
+      __atomic_store_n(&fail, -1, __ATOMIC_RELAXED);
+  }
+  return fail;
+}
+
 int m3x_oracle_bls_h2c_g2(const uint8_t msg[32], uint8_t out_uncomp[192]) {
   bls_init();
   g2_aff_t h;

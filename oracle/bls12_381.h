@@ -57,6 +57,12 @@ int m3x_oracle_bls_verify_sets(const uint8_t *msgs, const uint8_t *sigs,
                                const uint8_t *pks, const uint32_t *pk_offsets,
                                const uint64_t *rands, uint64_t n);
 
+
+/* workload-generation helpers (OpenMP batch; tests/bench only) */
+int m3x_oracle_bls_keypool(uint64_t n, uint8_t *sks, uint8_t *pks_uncomp);
+int m3x_oracle_bls_sign_batch(uint64_t n, const uint8_t *sks,
+                              const uint8_t *msgs, uint8_t *sigs);
+
 /* introspection for parity tests */
 int m3x_oracle_bls_h2c_g2(const uint8_t msg[32], uint8_t out_uncomp[192]);
 /* e(P,Q): P 96B uncomp G1, Q 192B uncomp G2; out 576B (12*48 BE,

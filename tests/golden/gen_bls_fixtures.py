@@ -807,7 +807,7 @@ def limbs(v, n=6):
 
 def c_limbs(name, v, n=6):
     ls = ", ".join(f"0x{w:016x}ULL" for w in limbs(v, n))
-    return f"static constexpr uint64_t {name}[{n}] = {{{ls}}};\n"
+    return f"static M3X_CONST uint64_t {name}[{n}] = {{{ls}}};\n"
 
 
 def c_fp2(name, v):
@@ -828,9 +828,10 @@ def emit_consts(path):
     out.append(" * Montgomery) form unless noted. */\n")
     out.append("#ifndef M3X_BLS_CONSTS_H\n#define M3X_BLS_CONSTS_H\n")
     out.append("#include <stdint.h>\n")
+    out.append("#ifdef __cplusplus\n#define M3X_CONST constexpr\n#else\n#define M3X_CONST const\n#endif\n")
     out.append(c_limbs("BLS_P", P))
     out.append(c_limbs("BLS_R2", R2))  # Montgomery R^2 mod p
-    out.append(f"static constexpr uint64_t BLS_N0 = 0x{N0:016x}ULL;\n")
+    out.append(f"static M3X_CONST uint64_t BLS_N0 = 0x{N0:016x}ULL;\n")
     out.append(c_limbs("BLS_ORDER", R, 4))
     out.append(f"#define BLS_X_ABS 0x{H_EFF_ABS:016x}ULL\n")
     out.append("#define BLS_X_NEGATIVE 1\n")
