@@ -912,6 +912,10 @@ __device__ inline void f12_one(fp12m &r) {
   fp_one(r.s[0]);
 }
 
+__device__ inline void f12_copy(fp12m &r, const fp12m &a) {
+  for (int i = 0; i < 12; i++) r.s[i] = a.s[i];
+}
+
 __device__ __forceinline__ void f12_get(const fp12m &a, int k, fp2 &c) {
   c.c0 = a.s[2 * k];
   c.c1 = a.s[2 * k + 1];
@@ -921,63 +925,64 @@ __device__ __forceinline__ void f12_set(fp12m &a, int k, const fp2 &c) {
   a.s[2 * k + 1] = c.c1;
 }
 
-__device__ inline void f12_mul(fp12m &r, const fp12m &a, const fp12m &b) {
-  fp12m lo, hi; // acc[0..5] in lo, acc[6..10] in hi slots 0..4
-  for (int i = 0; i < 12; i++) {
-    fp_zero(lo.s[i]);
-    fp_zero(hi.s[i]);
-  }
-  for (int i = 0; i < 6; i++) {
-    fp2 ai;
-    f12_get(a, i, ai);
-    if (fp2_is_zero(ai)) continue;
-    for (int j = 0; j < 6; j++) {
-      fp2 bj, t, cur;
-      f12_get(b, j, bj);
-      if (fp2_is_zero(bj)) continue;
-      fp2_mul(t, ai, bj);
-      int k = i + j;
-      if (k < 6) {
-        f12_get(lo, k, cur);
-        fp2_add(cur, cur, t);
-        f12_set(lo, k, cur);
-      } else {
-        f12_get(hi, k - 6, cur);
-        fp2_add(cur, cur, t);
-        f12_set(hi, k - 6, cur);
-      }
-    }
-  }
-  for (int k = 0; k < 5; k++) {
-    fp2 h, cur;
-    f12_get(hi, k, h);
-    fp2_mul_xi(h, h);
-    f12_get(lo, k, cur);
-    fp2_add(cur, cur, h);
-    f12_set(lo, k, cur);
-  }
-  r = lo;
-}
-
-__device__ inline void f12_sqr(fp12m &r, const fp12m &a) { f12_mul(r, a, a); }
-
-// multiply by the sparse line a0*w^0 + a3*w^3 + a5*w^5 (a0 in Fp2)
-__device__ inline void f12_mul_line(fp12m &f, const fp2 &a0, const fp2 &a3,
-                                    const fp2 &a5) {
-  fp12m l;
-  for (int i = 0; i < 12; i++) fp_zero(l.s[i]);
-  f12_set(l, 0, a0);
-  f12_set(l, 3, a3);
-  f12_set(l, 5, a5);
-  f12_mul(f, f, l);
-}
-
-__device__ inline void f12_conj6(fp12m &r, const fp12m &a) {
+// r = a * b with register accumulators; r must NOT alias a or b (the
+// non-aliasing form avoids the scratch-resident lo/hi temporaries that
+// dominated the first-cut kernel at low occupancy).
+__device__ inline void f12_mul_nn(fp12m &r, const fp12m &a, const fp12m &b) {
   for (int k = 0; k < 6; k++) {
+    fp2 acc, hacc;
+    fp2_zero(acc);
+    fp2_zero(hacc);
+    for (int i = 0; i <= k; i++) {
+      fp2 ai, bj, t;
+      f12_get(a, i, ai);
+      f12_get(b, k - i, bj);
+      fp2_mul(t, ai, bj);
+      fp2_add(acc, acc, t);
+    }
+    for (int i = k + 1; i < 6; i++) {
+      fp2 ai, bj, t;
+      f12_get(a, i, ai);
+      f12_get(b, k + 6 - i, bj);
+      fp2_mul(t, ai, bj);
+      fp2_add(hacc, hacc, t);
+    }
+    fp2_mul_xi(hacc, hacc);
+    fp2_add(acc, acc, hacc);
+    f12_set(r, k, acc);
+  }
+}
+
+__device__ inline void f12_sqr_nn(fp12m &r, const fp12m &a) {
+  f12_mul_nn(r, a, a);
+}
+
+// r = f * (a0 + a3 w^3 + a5 w^5); r must not alias f (sparse: 18 fp2 muls)
+__device__ inline void f12_line_nn(fp12m &r, const fp12m &f, const fp2 &a0,
+                                   const fp2 &a3, const fp2 &a5) {
+  for (int k = 0; k < 6; k++) {
+    fp2 acc, fk, t;
+    f12_get(f, k, fk);
+    fp2_mul(acc, fk, a0);
+    f12_get(f, (k + 3) % 6, fk);
+    fp2_mul(t, fk, a3);
+    if (k < 3) fp2_mul_xi(t, t);
+    fp2_add(acc, acc, t);
+    f12_get(f, (k + 1) % 6, fk); // (k - 5) mod 6
+    fp2_mul(t, fk, a5);
+    if (k < 5) fp2_mul_xi(t, t);
+    fp2_add(acc, acc, t);
+    f12_set(r, k, acc);
+  }
+}
+
+// f^(p^6): odd w-coefficients negate (in place; validated by the generator)
+__device__ inline void f12_conj6_ip(fp12m &a) {
+  for (int k = 1; k < 6; k += 2) {
     fp2 c;
     f12_get(a, k, c);
-    if (k & 1) fp2_neg(c, c);
-    f12_set(r, k, c);
+    fp2_neg(c, c);
+    f12_set(a, k, c);
   }
 }
 
@@ -990,53 +995,54 @@ __device__ inline bool f12_is_one(const fp12m &a) {
   return true;
 }
 
-// f^p: c_k -> conj(c_k) * FW1^k ; f^(p^2): c_k -> c_k * FW2^k
+// f^p: c_k -> conj(c_k)*FW1^k ; f^(p^2): c_k -> c_k*FW2^k (alias-safe)
 __device__ inline void f12_frob(fp12m &r, const fp12m &a, int power) {
   fp2 fw1, fw;
   FP2_LOAD_C(fw1, FROB_W1);
   if (power == 2) {
     fp2 c1;
     fp2_conj(c1, fw1);
-    fp2_mul(fw, fw1, c1); // FW2 = norm(FW1) in Fp
+    fp2_mul(fw, fw1, c1); // FW2 = norm(FW1), lands in Fp
   } else {
     fw = fw1;
   }
-  fp2 acc;
-  fp2_one(acc);
+  fp2 accw;
+  fp2_one(accw);
   for (int k = 0; k < 6; k++) {
     fp2 c;
     f12_get(a, k, c);
     if (power == 1) fp2_conj(c, c);
-    fp2_mul(c, c, acc);
+    fp2_mul(c, c, accw);
     f12_set(r, k, c);
-    fp2_mul(acc, acc, fw);
+    fp2_mul(accw, accw, fw);
   }
 }
 
-__device__ inline void f12_inv(fp12m &r, const fp12m &a) {
+// r = a^-1 via the sigma-conjugate product; r, g, t distinct from a and
+// from each other.
+__device__ inline void f12_inv_nn(fp12m &r, const fp12m &a, fp12m &g,
+                                  fp12m &t) {
   fp2 z6;
   FP2_LOAD_C(z6, ZETA6);
-  fp12m g, t;
   f12_one(g);
   for (int i = 1; i < 6; i++) {
-    fp2 zi, acc;
-    fp2_one(acc);
-    // zi = z6^i
+    fp2 zi, accw;
     zi = z6;
     for (int q = 1; q < i; q++) fp2_mul(zi, zi, z6);
+    fp2_one(accw);
     for (int j = 0; j < 6; j++) {
       fp2 c;
       f12_get(a, j, c);
-      fp2_mul(c, c, acc);
+      fp2_mul(c, c, accw);
       f12_set(t, j, c);
-      fp2_mul(acc, acc, zi);
+      fp2_mul(accw, accw, zi);
     }
-    f12_mul(g, g, t);
+    f12_mul_nn(r, g, t);
+    f12_copy(g, r);
   }
-  fp12m n;
-  f12_mul(n, a, g);
+  f12_mul_nn(t, a, g); // norm (lands in Fp2)
   fp2 n0, ninv;
-  f12_get(n, 0, n0);
+  f12_get(t, 0, n0);
   fp2_inv(ninv, n0);
   for (int j = 0; j < 6; j++) {
     fp2 c;
@@ -1046,41 +1052,48 @@ __device__ inline void f12_inv(fp12m &r, const fp12m &a) {
   }
 }
 
-// f^(|x|) with |x| = BLS_X_ABS (hamming weight 6); in the cyclotomic
-// subgroup f^-1 = conj6(f), so f^x (x negative) = conj6(f^|x|).
-__device__ inline void f12_pow_xabs(fp12m &r, const fp12m &a) {
-  fp12m acc = a;
+// r = a^|x| (|x| = BLS_X_ABS, hamming 6); r, t distinct from a and each
+// other. In the cyclotomic subgroup a^x (x<0) = conj6(a^|x|).
+__device__ inline void f12_pow_xabs_nn(fp12m &r, const fp12m &a, fp12m &t) {
+  f12_copy(r, a);
   for (int b = 62; b >= 0; b--) {
-    f12_sqr(acc, acc);
-    if ((BLS_X_ABS >> b) & 1) f12_mul(acc, acc, a);
+    f12_sqr_nn(t, r);
+    if ((BLS_X_ABS >> b) & 1)
+      f12_mul_nn(r, t, a);
+    else
+      f12_copy(r, t);
   }
-  r = acc;
 }
 
 // ---------------------------------------------------------------- pairing ---
 
-// f *= miller(P, Q): inversion-free Jacobian loop on the twist with sparse
-// lines (validated formulas; see header comment).
-__device__ inline void miller_acc(fp12m &f, const g1a &Pa, const g2a &Qa) {
-  if (Pa.inf || Qa.inf) return;
+// out = miller(P, Q) (already conjugated for x<0); tmp is scratch; out and
+// tmp must be distinct. Inversion-free Jacobian loop on the twist with
+// sparse lines (formulas validated against the Python reference).
+__device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
+                                  const g2a &Qa) {
+  f12_one(out);
+  if (Pa.inf || Qa.inf) return; // e(O,.) = e(.,O) = 1
   g2j T;
   g2j_from_aff(T, Qa);
   fp2 xi_inv;
   {
-    fp2 xi;
-    fp2 one;
+    fp2 xi, one;
     fp2_one(one);
     fp2_mul_xi(xi, one);
     fp2_inv(xi_inv, xi);
   }
-  fp12m acc;
-  f12_one(acc);
+  fp12m *cur = &out, *tm = &tmp;
   fp xp = Pa.x, yp = Pa.y;
   for (int i = 62; i >= 0; i--) {
-    f12_sqr(acc, acc);
+    f12_sqr_nn(*tm, *cur);
+    {
+      fp12m *sw = cur;
+      cur = tm;
+      tm = sw;
+    }
     // doubling line from Jacobian T:
-    // a0 = 2*Y*Z^3 * yp ; a3 = (3X^3 - 2Y^2)*xi_inv ;
-    // a5 = -3X^2*Z^2 * xp * xi_inv
+    // a0 = 2*Y*Z^3*yp ; a3 = (3X^3 - 2Y^2)*xi_inv ; a5 = -3X^2*Z^2*xp*xi_inv
     {
       fp2 X2, Y2, Z2, Z3, a0, a3, a5, t, t2;
       fp2_sqr(X2, T.x);
@@ -1100,13 +1113,16 @@ __device__ inline void miller_acc(fp12m &f, const g1a &Pa, const g2a &Qa) {
       fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
-      f12_mul_line(acc, a0, a3, a5);
+      f12_line_nn(*tm, *cur, a0, a3, a5);
+      {
+        fp12m *sw = cur;
+        cur = tm;
+        tm = sw;
+      }
       g2j_dbl(T, T);
     }
     if ((BLS_X_ABS >> i) & 1) {
-      // addition line through T (jac) and Q (affine):
-      // H = X - xQ*Z^2, M = Y - yQ*Z^3
-      // a0 = Z^3*H*yp ; a3 = (M*X - Y*H)*xi_inv ; a5 = -M*Z^2*xp*xi_inv
+      // addition line through T (jac) and Q (affine)
       fp2 Z2, Z3, H, M, a0, a3, a5, t, t2;
       fp2_sqr(Z2, T.z);
       fp2_mul(Z3, Z2, T.z);
@@ -1124,56 +1140,59 @@ __device__ inline void miller_acc(fp12m &f, const g1a &Pa, const g2a &Qa) {
       fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
-      f12_mul_line(acc, a0, a3, a5);
-      // mixed add T += Q
+      f12_line_nn(*tm, *cur, a0, a3, a5);
+      {
+        fp12m *sw = cur;
+        cur = tm;
+        tm = sw;
+      }
       g2j qj;
       g2j_from_aff(qj, Qa);
       g2j_add(T, T, qj);
     }
   }
-  fp12m conj;
-  f12_conj6(conj, acc); // x < 0
-  f12_mul(f, f, conj);
+  f12_conj6_ip(*cur); // x < 0
+  if (cur != &out) f12_copy(out, *cur);
 }
 
 // final exponentiation: easy part + HHT chain
 // hard = (x-1)^2 (x+p) (x^2+p^2-1) + 3  (== 3*(p^4-p^2+1)/r; the cube is
-// harmless for the ==1 test since gcd(3,r)=1 — validated in Python)
-__device__ inline void final_exp(fp12m &r, const fp12m &f) {
-  fp12m t, fi, e;
-  f12_conj6(t, f);
-  f12_inv(fi, f);
-  f12_mul(e, t, fi); // f^(p^6-1)
-  f12_frob(t, e, 2);
-  f12_mul(e, t, e); // ^(p^2+1) — now in the cyclotomic subgroup
-  // u = e^((x-1)^2): e^(x-1) twice, with x negative:
-  // e^(x-1) = conj6(e^(|x|+1))
-  fp12m u, v;
-  f12_pow_xabs(u, e);
-  f12_mul(u, u, e);
-  f12_conj6(u, u); // e^(x-1)
-  f12_pow_xabs(v, u);
-  f12_mul(v, v, u);
-  f12_conj6(v, v); // e^((x-1)^2)
-  // w1 = v^(x+p) = conj6(v^|x|) * v^p
-  fp12m w1, w2;
-  f12_pow_xabs(w1, v);
-  f12_conj6(w1, w1);
-  f12_frob(t, v, 1);
-  f12_mul(w1, w1, t);
-  // w2 = w1^(x^2+p^2-1) = (w1^x)^x * w1^(p^2) * conj6(w1)
-  f12_pow_xabs(w2, w1);
-  f12_conj6(w2, w2);
-  f12_pow_xabs(w2, w2);
-  f12_conj6(w2, w2); // w1^(x^2)
-  f12_frob(t, w1, 2);
-  f12_mul(w2, w2, t);
-  f12_conj6(t, w1);
-  f12_mul(w2, w2, t);
-  // result = w2 * e^3
-  f12_sqr(t, e);
-  f12_mul(t, t, e);
-  f12_mul(r, w2, t);
+// harmless for the ==1 test since gcd(3,r)=1 — validated in Python).
+// r and f must be distinct from each other and from the 6 scratch slots.
+__device__ inline void final_exp_s(fp12m &r, const fp12m &f, fp12m *s) {
+  f12_copy(s[0], f);
+  f12_conj6_ip(s[0]); // f^(p^6)
+  f12_inv_nn(s[1], f, s[2], s[3]);
+  f12_mul_nn(s[2], s[0], s[1]); // f^(p^6-1)
+  f12_frob(s[0], s[2], 2);
+  f12_mul_nn(s[1], s[0], s[2]); // e = f^((p^6-1)(p^2+1)), cyclotomic
+  // u = e^(x-1) = conj6(e^|x| * e)
+  f12_pow_xabs_nn(s[0], s[1], s[3]);
+  f12_mul_nn(s[2], s[0], s[1]);
+  f12_conj6_ip(s[2]); // e^(x-1)
+  // v = u^(x-1)
+  f12_pow_xabs_nn(s[0], s[2], s[3]);
+  f12_mul_nn(s[4], s[0], s[2]);
+  f12_conj6_ip(s[4]); // e^((x-1)^2)
+  // w1 = conj6(v^|x|) * v^p
+  f12_pow_xabs_nn(s[0], s[4], s[3]);
+  f12_conj6_ip(s[0]);
+  f12_frob(s[2], s[4], 1);
+  f12_mul_nn(s[5], s[0], s[2]); // w1
+  // w2 = (w1^x)^x * w1^(p^2) * conj6(w1)
+  f12_pow_xabs_nn(s[0], s[5], s[3]);
+  f12_conj6_ip(s[0]);
+  f12_pow_xabs_nn(s[2], s[0], s[3]);
+  f12_conj6_ip(s[2]); // w1^(x^2)
+  f12_frob(s[0], s[5], 2);
+  f12_mul_nn(s[4], s[2], s[0]);
+  f12_copy(s[0], s[5]);
+  f12_conj6_ip(s[0]);
+  f12_mul_nn(s[2], s[4], s[0]); // w2
+  // r = w2 * e^3 (e still in s[1])
+  f12_sqr_nn(s[0], s[1]);
+  f12_mul_nn(s[3], s[0], s[1]);
+  f12_mul_nn(r, s[2], s[3]);
 }
 
 // --------------------------------------------------------- hash-to-curve ---

@@ -22,8 +22,10 @@ struct BlsWork {
   g2j *rsig;       // [n] r_i * sigma (jacobian)
   fp12m *fparts;   // [n] per-set miller values
   int *fail;       // [1]
-  g2j *sig_sum;    // [1 + 256 scratch]
-  fp12m *gt_parts; // [256]
+  g2j *sig_stage;  // [256] stage-1 partial sums
+  g2j *sig_sum;    // [1]
+  fp12m *gt_stage; // [256] stage-1 partial products
+  fp12m *gt_parts; // [1]
   int *verdict;    // [1]
 };
 
@@ -126,39 +128,54 @@ __global__ void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
 __global__ void k_bls_miller(uint64_t n, BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  fp12m f;
-  f12_one(f);
-  if (*w.fail == 0) miller_acc(f, w.p_scaled[i], w.h2c[i]);
+  fp12m f, tmp;
+  if (*w.fail == 0)
+    miller_raw(f, tmp, w.p_scaled[i], w.h2c[i]);
+  else
+    f12_one(f);
   w.fparts[i] = f;
 }
 
-// single 256-thread block: product of all per-set miller values via LDS tree
-__global__ __launch_bounds__(256) void k_bls_reduce_gt(uint64_t n, BlsWork w) {
+// two-stage GT-product reduction: each block folds its contiguous span of
+// per-set miller values (thread-strided local products, then an LDS tree)
+// into one output element; a second 1-block launch folds the partials.
+__global__ __launch_bounds__(256) void k_bls_reduce_gt(
+    const fp12m *__restrict__ in, uint64_t n, fp12m *__restrict__ out) {
   __shared__ fp12m lds[256];
-  fp12m local;
+  uint64_t per = (n + gridDim.x - 1) / gridDim.x;
+  uint64_t lo = (uint64_t)blockIdx.x * per;
+  uint64_t hi = lo + per < n ? lo + per : n;
+  fp12m local, t;
   f12_one(local);
-  for (uint64_t i = threadIdx.x; i < n; i += 256) f12_mul(local, local, w.fparts[i]);
+  for (uint64_t i = lo + threadIdx.x; i < hi; i += 256) {
+    f12_mul_nn(t, local, in[i]);
+    f12_copy(local, t);
+  }
   lds[threadIdx.x] = local;
   __syncthreads();
   for (int s = 128; s > 0; s >>= 1) {
     if ((int)threadIdx.x < s) {
-      fp12m t;
-      f12_mul(t, lds[threadIdx.x], lds[threadIdx.x + s]);
-      lds[threadIdx.x] = t;
+      f12_mul_nn(t, lds[threadIdx.x], lds[threadIdx.x + s]);
+      f12_copy(lds[threadIdx.x], t);
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) w.gt_parts[0] = lds[0];
+  if (threadIdx.x == 0) out[blockIdx.x] = lds[0];
 }
 
-// single 256-thread block: sum of all r_i*sigma_i via LDS tree
-__global__ __launch_bounds__(256) void k_bls_reduce_sig(uint64_t n, BlsWork w) {
+// two-stage sum of r_i*sigma_i, same shape
+__global__ __launch_bounds__(256) void k_bls_reduce_sig(
+    const g2j *__restrict__ in, uint64_t n, g2j *__restrict__ out) {
   __shared__ g2j lds[256];
+  uint64_t per = (n + gridDim.x - 1) / gridDim.x;
+  uint64_t lo = (uint64_t)blockIdx.x * per;
+  uint64_t hi = lo + per < n ? lo + per : n;
   g2j local;
   fp2_zero(local.x);
   fp2_zero(local.y);
   fp2_zero(local.z);
-  for (uint64_t i = threadIdx.x; i < n; i += 256) g2j_add(local, local, w.rsig[i]);
+  for (uint64_t i = lo + threadIdx.x; i < hi; i += 256)
+    g2j_add(local, local, in[i]);
   lds[threadIdx.x] = local;
   __syncthreads();
   for (int s = 128; s > 0; s >>= 1) {
@@ -169,26 +186,29 @@ __global__ __launch_bounds__(256) void k_bls_reduce_sig(uint64_t n, BlsWork w) {
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) w.sig_sum[0] = lds[0];
+  if (threadIdx.x == 0) out[blockIdx.x] = lds[0];
 }
 
-// final: f_total *= miller(-g1, sig_sum); final_exp; compare to one
+// final: f_total *= miller(-g1, sig_sum); final_exp; compare to one.
+// Single-thread control flow but the Fp12 working set lives in LDS (the
+// scratch-latency fix for the one-per-batch serial tail).
 __global__ void k_bls_finish(BlsWork w) {
+  __shared__ fp12m sh[9];
   if (threadIdx.x != 0 || blockIdx.x != 0) return;
   if (*w.fail) {
     *w.verdict = 0;
     return;
   }
-  fp12m f = w.gt_parts[0];
   g2a sig_a;
   g2j_to_aff(sig_a, w.sig_sum[0]);
   g1a ng1;
   g1_gen(ng1);
   fp_neg(ng1.y, ng1.y);
-  miller_acc(f, ng1, sig_a);
-  fp12m e;
-  final_exp(e, f);
-  *w.verdict = f12_is_one(e) ? 1 : 0;
+  f12_copy(sh[0], w.gt_parts[0]);
+  miller_raw(sh[1], sh[2], ng1, sig_a);
+  f12_mul_nn(sh[2], sh[0], sh[1]); // f_total
+  final_exp_s(sh[1], sh[2], &sh[3]);
+  *w.verdict = f12_is_one(sh[1]) ? 1 : 0;
 }
 
 int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
@@ -202,7 +222,9 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_f = bytes; bytes += align(n * sizeof(fp12m));
   uint64_t off_fail = bytes; bytes += 256;
+  uint64_t off_sig_stage = bytes; bytes += align(256 * sizeof(g2j));
   uint64_t off_sum = bytes; bytes += align(sizeof(g2j));
+  uint64_t off_gt_stage = bytes; bytes += align(256 * sizeof(fp12m));
   uint64_t off_gt = bytes; bytes += align(sizeof(fp12m));
   uint64_t off_v = bytes; bytes += 256;
   int rc = m3x::ensure_scratch(ctx, &ctx->scratch_a, &ctx->scratch_a_bytes,
@@ -214,7 +236,9 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.rsig = reinterpret_cast<g2j *>(base + off_r);
   w.fparts = reinterpret_cast<fp12m *>(base + off_f);
   w.fail = reinterpret_cast<int *>(base + off_fail);
+  w.sig_stage = reinterpret_cast<g2j *>(base + off_sig_stage);
   w.sig_sum = reinterpret_cast<g2j *>(base + off_sum);
+  w.gt_stage = reinterpret_cast<fp12m *>(base + off_gt_stage);
   w.gt_parts = reinterpret_cast<fp12m *>(base + off_gt);
   w.verdict = reinterpret_cast<int *>(base + off_v);
   M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
@@ -233,11 +257,17 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
                      w);
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
+  uint32_t rblocks = (uint32_t)((n + 255) / 256);
+  if (rblocks > 256) rblocks = 256;
   m3x::time_begin(ctx, M3X_K_BLS_REDUCE);
-  hipLaunchKernelGGL(k_bls_reduce_gt, dim3(1), dim3(256), 0, ctx->stream, n,
-                     w);
-  hipLaunchKernelGGL(k_bls_reduce_sig, dim3(1), dim3(256), 0, ctx->stream, n,
-                     w);
+  hipLaunchKernelGGL(k_bls_reduce_gt, dim3(rblocks), dim3(256), 0,
+                     ctx->stream, w.fparts, n, w.gt_stage);
+  hipLaunchKernelGGL(k_bls_reduce_gt, dim3(1), dim3(256), 0, ctx->stream,
+                     w.gt_stage, (uint64_t)rblocks, w.gt_parts);
+  hipLaunchKernelGGL(k_bls_reduce_sig, dim3(rblocks), dim3(256), 0,
+                     ctx->stream, w.rsig, n, w.sig_stage);
+  hipLaunchKernelGGL(k_bls_reduce_sig, dim3(1), dim3(256), 0, ctx->stream,
+                     w.sig_stage, (uint64_t)rblocks, w.sig_sum);
   m3x::time_end(ctx, M3X_K_BLS_REDUCE);
   m3x::time_begin(ctx, M3X_K_BLS_FINISH);
   hipLaunchKernelGGL(k_bls_finish, dim3(1), dim3(64), 0, ctx->stream, w);
