@@ -1195,6 +1195,265 @@ __device__ inline void final_exp_s(fp12m &r, const fp12m &f, fp12m *s) {
   f12_mul_nn(r, s[2], s[3]);
 }
 
+// ------------------------- wave-cooperative fp12 (64-thread finish path) ---
+// All 64 lanes execute the same control flow; fp12 multiplies fan the 36
+// coefficient products across lanes through an LDS staging area. Ops are
+// ALIAS-SAFE (reads complete before the sync that precedes writes), so
+// squarings and accumulations run in place. Requires blockDim.x == 64.
+
+struct f12w_ws {
+  fp2 t[36];
+};
+
+__device__ __forceinline__ void f12w_sync() { __syncthreads(); }
+
+__device__ inline void f12_mul_w(fp12m &r, const fp12m &a, const fp12m &b,
+                                 f12w_ws &ws, int lane) {
+  if (lane < 36) {
+    int i = lane / 6, j = lane % 6;
+    fp2 ai, bj, t;
+    f12_get(a, i, ai);
+    f12_get(b, j, bj);
+    fp2_mul(t, ai, bj);
+    ws.t[lane] = t;
+  }
+  f12w_sync();
+  if (lane < 6) {
+    fp2 acc, h, t;
+    fp2_zero(acc);
+    fp2_zero(h);
+    for (int i = 0; i <= lane; i++) {
+      t = ws.t[i * 6 + (lane - i)];
+      fp2_add(acc, acc, t);
+    }
+    for (int i = lane + 1; i < 6; i++) {
+      t = ws.t[i * 6 + (lane + 6 - i)];
+      fp2_add(h, h, t);
+    }
+    fp2_mul_xi(h, h);
+    fp2_add(acc, acc, h);
+    f12_set(r, lane, acc);
+  }
+  f12w_sync();
+}
+
+// f *= (a0 + a3 w^3 + a5 w^5), in place
+__device__ inline void f12_line_w(fp12m &f, const fp2 &a0, const fp2 &a3,
+                                  const fp2 &a5, f12w_ws &ws, int lane) {
+  if (lane < 18) {
+    int k = lane / 3, q = lane % 3;
+    fp2 fk, t;
+    if (q == 0) {
+      f12_get(f, k, fk);
+      fp2_mul(t, fk, a0);
+    } else if (q == 1) {
+      f12_get(f, (k + 3) % 6, fk);
+      fp2_mul(t, fk, a3);
+      if (k < 3) fp2_mul_xi(t, t);
+    } else {
+      f12_get(f, (k + 1) % 6, fk);
+      fp2_mul(t, fk, a5);
+      if (k < 5) fp2_mul_xi(t, t);
+    }
+    ws.t[lane] = t;
+  }
+  f12w_sync();
+  if (lane < 6) {
+    fp2 acc, t;
+    acc = ws.t[3 * lane];
+    t = ws.t[3 * lane + 1];
+    fp2_add(acc, acc, t);
+    t = ws.t[3 * lane + 2];
+    fp2_add(acc, acc, t);
+    f12_set(f, lane, acc);
+  }
+  f12w_sync();
+}
+
+__device__ inline void f12_copy_w(fp12m &r, const fp12m &a, int lane) {
+  if (lane < 12) r.s[lane] = a.s[lane];
+  f12w_sync();
+}
+
+__device__ inline void f12_conj6_w(fp12m &a, int lane) {
+  if (lane < 3) {
+    fp2 c;
+    f12_get(a, 2 * lane + 1, c);
+    fp2_neg(c, c);
+    f12_set(a, 2 * lane + 1, c);
+  }
+  f12w_sync();
+}
+
+__device__ inline void f12_frob_w(fp12m &r, const fp12m &a, int power,
+                                  int lane) {
+  if (lane < 6) {
+    fp2 fw1, fw, accw, c;
+    FP2_LOAD_C(fw1, FROB_W1);
+    if (power == 2) {
+      fp2 c1;
+      fp2_conj(c1, fw1);
+      fp2_mul(fw, fw1, c1);
+    } else {
+      fw = fw1;
+    }
+    fp2_one(accw);
+    for (int q = 0; q < lane; q++) fp2_mul(accw, accw, fw);
+    f12_get(a, lane, c);
+    if (power == 1) fp2_conj(c, c);
+    fp2_mul(c, c, accw);
+    f12_set(r, lane, c);
+  }
+  f12w_sync();
+}
+
+// r = a^-1; r, g, t distinct from a and each other
+__device__ inline void f12_inv_w(fp12m &r, const fp12m &a, fp12m &g, fp12m &t,
+                                 f12w_ws &ws, int lane) {
+  if (lane == 0) f12_one(g);
+  f12w_sync();
+  for (int i = 1; i < 6; i++) {
+    if (lane < 6) {
+      fp2 z6, zi, fac, c;
+      FP2_LOAD_C(z6, ZETA6);
+      fp2_one(zi);
+      for (int q = 0; q < i; q++) fp2_mul(zi, zi, z6);
+      fp2_one(fac);
+      for (int q = 0; q < lane; q++) fp2_mul(fac, fac, zi);
+      f12_get(a, lane, c);
+      fp2_mul(c, c, fac);
+      f12_set(t, lane, c);
+    }
+    f12w_sync();
+    f12_mul_w(g, g, t, ws, lane); // in-place safe
+  }
+  f12_mul_w(t, a, g, ws, lane); // norm in t.c0
+  if (lane == 0) {
+    fp2 n0, ninv;
+    f12_get(t, 0, n0);
+    fp2_inv(ninv, n0);
+    ws.t[0] = ninv;
+  }
+  f12w_sync();
+  if (lane < 6) {
+    fp2 c, ninv = ws.t[0];
+    f12_get(g, lane, c);
+    fp2_mul(c, c, ninv);
+    f12_set(r, lane, c);
+  }
+  f12w_sync();
+}
+
+// r = a^|x|; r distinct from a
+__device__ inline void f12_pow_xabs_w(fp12m &r, const fp12m &a, f12w_ws &ws,
+                                      int lane) {
+  f12_copy_w(r, a, lane);
+  for (int b = 62; b >= 0; b--) {
+    f12_mul_w(r, r, r, ws, lane);
+    if ((BLS_X_ABS >> b) & 1) f12_mul_w(r, r, a, ws, lane);
+  }
+}
+
+// out = miller(P, Q) (conjugated); in-place cooperative version
+__device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
+                                f12w_ws &ws, int lane) {
+  if (lane == 0) f12_one(out);
+  f12w_sync();
+  if (Pa.inf || Qa.inf) return;
+  g2j T;
+  g2j_from_aff(T, Qa);
+  fp2 xi_inv;
+  {
+    fp2 xi, one;
+    fp2_one(one);
+    fp2_mul_xi(xi, one);
+    fp2_inv(xi_inv, xi);
+  }
+  fp xp = Pa.x, yp = Pa.y;
+  for (int i = 62; i >= 0; i--) {
+    f12_mul_w(out, out, out, ws, lane);
+    {
+      fp2 X2, Y2, Z2, Z3, a0, a3, a5, t, t2;
+      fp2_sqr(X2, T.x);
+      fp2_sqr(Y2, T.y);
+      fp2_sqr(Z2, T.z);
+      fp2_mul(Z3, Z2, T.z);
+      fp2_mul(t, T.y, Z3);
+      fp2_dbl(t, t);
+      fp2_mul_fp(a0, t, yp);
+      fp2_mul(t, X2, T.x);
+      fp2_mul_small(t, t, 3);
+      fp2_dbl(t2, Y2);
+      fp2_sub(t, t, t2);
+      fp2_mul(a3, t, xi_inv);
+      fp2_mul(t, X2, Z2);
+      fp2_mul_small(t, t, 3);
+      fp2_mul_fp(t, t, xp);
+      fp2_neg(t, t);
+      fp2_mul(a5, t, xi_inv);
+      f12_line_w(out, a0, a3, a5, ws, lane);
+      g2j_dbl(T, T);
+    }
+    if ((BLS_X_ABS >> i) & 1) {
+      fp2 Z2, Z3, H, M, a0, a3, a5, t, t2;
+      fp2_sqr(Z2, T.z);
+      fp2_mul(Z3, Z2, T.z);
+      fp2_mul(t, Qa.x, Z2);
+      fp2_sub(H, T.x, t);
+      fp2_mul(t, Qa.y, Z3);
+      fp2_sub(M, T.y, t);
+      fp2_mul(t, Z3, H);
+      fp2_mul_fp(a0, t, yp);
+      fp2_mul(t, M, T.x);
+      fp2_mul(t2, T.y, H);
+      fp2_sub(t, t, t2);
+      fp2_mul(a3, t, xi_inv);
+      fp2_mul(t, M, Z2);
+      fp2_mul_fp(t, t, xp);
+      fp2_neg(t, t);
+      fp2_mul(a5, t, xi_inv);
+      f12_line_w(out, a0, a3, a5, ws, lane);
+      g2j qj;
+      g2j_from_aff(qj, Qa);
+      g2j_add(T, T, qj);
+    }
+  }
+  f12_conj6_w(out, lane);
+}
+
+// r = final_exp(f); r, f distinct from the 4 scratch slots s[0..3]
+__device__ inline void final_exp_w(fp12m &r, const fp12m &f, fp12m *s,
+                                   f12w_ws &ws, int lane) {
+  f12_copy_w(s[0], f, lane);
+  f12_conj6_w(s[0], lane); // f^(p^6)
+  f12_inv_w(s[1], f, s[2], s[3], ws, lane);
+  f12_mul_w(s[1], s[0], s[1], ws, lane); // f^(p^6-1)
+  f12_frob_w(s[0], s[1], 2, lane);
+  f12_mul_w(s[1], s[0], s[1], ws, lane); // e (cyclotomic)
+  f12_pow_xabs_w(s[0], s[1], ws, lane);  // e^|x|
+  f12_mul_w(s[0], s[0], s[1], ws, lane);
+  f12_conj6_w(s[0], lane); // u = e^(x-1)
+  f12_pow_xabs_w(s[2], s[0], ws, lane);
+  f12_mul_w(s[2], s[2], s[0], ws, lane);
+  f12_conj6_w(s[2], lane); // v = e^((x-1)^2)
+  f12_pow_xabs_w(s[3], s[2], ws, lane);
+  f12_conj6_w(s[3], lane); // v^x
+  f12_frob_w(s[0], s[2], 1, lane);
+  f12_mul_w(s[3], s[3], s[0], ws, lane); // w1 = v^(x+p)
+  f12_pow_xabs_w(s[0], s[3], ws, lane);
+  f12_conj6_w(s[0], lane); // w1^x
+  f12_pow_xabs_w(s[2], s[0], ws, lane);
+  f12_conj6_w(s[2], lane); // w1^(x^2)
+  f12_frob_w(s[0], s[3], 2, lane);
+  f12_mul_w(s[2], s[2], s[0], ws, lane);
+  f12_copy_w(s[0], s[3], lane);
+  f12_conj6_w(s[0], lane);
+  f12_mul_w(s[2], s[2], s[0], ws, lane); // w2
+  f12_mul_w(s[0], s[1], s[1], ws, lane); // e^2
+  f12_mul_w(s[0], s[0], s[1], ws, lane); // e^3
+  f12_mul_w(r, s[2], s[0], ws, lane);
+}
+
 // --------------------------------------------------------- hash-to-curve ---
 
 __device__ inline void h2f_from_be64(fp &r, const uint8_t *b) {

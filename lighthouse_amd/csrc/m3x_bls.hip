@@ -40,7 +40,7 @@ __device__ void order_be_bytes(uint8_t be[32]) {
 
 // ---------------------------------------------------------------- kernels
 
-__global__ void k_bls_pk_decompress(const uint8_t *__restrict__ comp,
+__global__ __launch_bounds__(64, 1) void k_bls_pk_decompress(const uint8_t *__restrict__ comp,
                                     uint64_t n, uint8_t *__restrict__ uncomp,
                                     int32_t *__restrict__ status) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -68,7 +68,7 @@ __global__ void k_bls_pk_decompress(const uint8_t *__restrict__ comp,
 
 // per-set prepare: sig decompress + subgroup check, pk aggregation,
 // r-scaling of both sides
-__global__ void k_bls_prepare(const uint8_t *__restrict__ sigs,
+__global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict__ sigs,
                               const uint8_t *__restrict__ pks,
                               const uint32_t *__restrict__ offs,
                               const uint64_t *__restrict__ rands, uint64_t n,
@@ -118,14 +118,14 @@ __global__ void k_bls_prepare(const uint8_t *__restrict__ sigs,
   }
 }
 
-__global__ void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
+__global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
                           BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   m3xb::h2c_g2(w.h2c[i], msgs + 32 * i);
 }
 
-__global__ void k_bls_miller(uint64_t n, BlsWork w) {
+__global__ __launch_bounds__(64, 1) void k_bls_miller(uint64_t n, BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   fp12m f, tmp;
@@ -190,25 +190,27 @@ __global__ __launch_bounds__(256) void k_bls_reduce_sig(
 }
 
 // final: f_total *= miller(-g1, sig_sum); final_exp; compare to one.
-// Single-thread control flow but the Fp12 working set lives in LDS (the
-// scratch-latency fix for the one-per-batch serial tail).
-__global__ void k_bls_finish(BlsWork w) {
-  __shared__ fp12m sh[9];
-  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+// One wave, cooperative fp12 ops in LDS (the per-batch serial tail: the
+// 36 coefficient products of each Fp12 multiply fan across lanes).
+__global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
+  __shared__ fp12m sh[7];
+  __shared__ f12w_ws ws;
+  int lane = threadIdx.x;
   if (*w.fail) {
-    *w.verdict = 0;
+    if (lane == 0) *w.verdict = 0;
     return;
   }
-  g2a sig_a;
+  g2a sig_a; // all lanes redundantly (wave-uniform)
   g2j_to_aff(sig_a, w.sig_sum[0]);
   g1a ng1;
   g1_gen(ng1);
   fp_neg(ng1.y, ng1.y);
-  f12_copy(sh[0], w.gt_parts[0]);
-  miller_raw(sh[1], sh[2], ng1, sig_a);
-  f12_mul_nn(sh[2], sh[0], sh[1]); // f_total
-  final_exp_s(sh[1], sh[2], &sh[3]);
-  *w.verdict = f12_is_one(sh[1]) ? 1 : 0;
+  if (lane == 0) f12_copy(sh[0], w.gt_parts[0]);
+  f12w_sync();
+  miller_w(sh[1], ng1, sig_a, ws, lane);
+  f12_mul_w(sh[1], sh[0], sh[1], ws, lane); // f_total
+  final_exp_w(sh[2], sh[1], &sh[3], ws, lane);
+  if (lane == 0) *w.verdict = f12_is_one(sh[2]) ? 1 : 0;
 }
 
 int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
