@@ -413,13 +413,7 @@ __device__ inline bool fp2_sqrt(fp2 &r, const fp2 &a) {
   fp_sqr(t, a.c1);
   fp_add(n, n, t);
   if (!fp_sqrt(s, n)) return false;
-  // 1/2 = (p+1)/2 mod p: compute as inv of mont(2)
-  {
-    fp two, one;
-    fp_one(one);
-    fp_add(two, one, one);
-    fp_inv(inv2, two);
-  }
+  FP_LOAD_C(inv2, FP_TWO_INV); // constant 2^-1 (generator-emitted)
   fp_add(d, a.c0, s);
   fp_mul(d, d, inv2);
   if (!fp_sqrt(x0, d)) {
@@ -1541,18 +1535,13 @@ __device__ inline void sswu_g2(g2a &out, const fp2 &u) {
   fp2_sqr(tv, zu2);
   fp2_add(tv, tv, zu2);
   if (fp2_is_zero(tv)) {
-    fp2 za;
-    fp2_mul(za, Z, A);
-    fp2_inv(za, za);
-    fp2_mul(x1, B, za);
+    FP2_LOAD_C(x1, SSWU_B_DIV_ZA); // constant B/(Z*A)
   } else {
     fp2 inv_tv, one, nb_over_a;
     fp2_inv(inv_tv, tv);
     fp2_one(one);
     fp2_add(inv_tv, one, inv_tv);
-    fp2_inv(nb_over_a, A);
-    fp2_mul(nb_over_a, nb_over_a, B);
-    fp2_neg(nb_over_a, nb_over_a);
+    FP2_LOAD_C(nb_over_a, SSWU_NB_DIV_A); // constant -B/A
     fp2_mul(x1, nb_over_a, inv_tv);
   }
   fp2_sqr(gx1, x1);
@@ -1619,10 +1608,15 @@ __device__ inline void iso_map_g2(g2a &out, const g2a &in) {
     fp2_mul(yd, yd, in.x);
     fp2_add(yd, yd, k[i]);
   }
-  fp2_inv(t, xd);
-  fp2_mul(out.x, xn, t);
-  fp2_inv(t, yd);
-  fp2_mul(out.y, yn, t);
+  // single shared inversion: 1/(xd*yd), then multiply back
+  fp2 prod;
+  fp2_mul(prod, xd, yd);
+  fp2_inv(t, prod);
+  fp2 xdi, ydi;
+  fp2_mul(ydi, t, xd); // 1/yd
+  fp2_mul(xdi, t, yd); // 1/xd
+  fp2_mul(out.x, xn, xdi);
+  fp2_mul(out.y, yn, ydi);
   fp2_mul(out.y, out.y, in.y);
   out.inf = 0;
 }

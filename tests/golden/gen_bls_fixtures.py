@@ -842,6 +842,12 @@ def emit_consts(path):
     out.append(c_fp2("SSWU_A", A_P))
     out.append(c_fp2("SSWU_B", B_P))
     out.append(c_fp2("SSWU_Z", Z_SSWU))
+    # derived constants for the GPU path (avoid per-call field inversions)
+    nb_div_a = f2mul(f2neg(B_P), f2inv(A_P))
+    b_div_za = f2mul(B_P, f2inv(f2mul(Z_SSWU, A_P)))
+    out.append(c_fp2("SSWU_NB_DIV_A", nb_div_a))
+    out.append(c_fp2("SSWU_B_DIV_ZA", b_div_za))
+    out.append(c_limbs("FP_TWO_INV", pow(2, P - 2, P)))
     for nm, arr in [
         ("ISO_XNUM", ISO_XNUM),
         ("ISO_XDEN", ISO_XDEN),
