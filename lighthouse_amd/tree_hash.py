@@ -57,6 +57,50 @@ def basic_vector_root(data: bytes, n_elems: int, elem_size: int, ctx=None) -> by
                             ctx=ctx)
 
 
+def merkleize_chunks_dev(dev_ptr, n_chunks: int, depth: int, mix_len: int = -1,
+                         ctx=None) -> bytes:
+    """device-resident chunk buffer (zero-padded tail chunk)."""
+    import ctypes as ct
+
+    ctx = ctx or _native.default_ctx()
+    out = ct.create_string_buffer(32)
+    rc = ctx._lib.m3x_merkleize_chunks_dev(ctx.handle, dev_ptr, n_chunks,
+                                           depth, mix_len, out)
+    if rc != 0:
+        raise RuntimeError(f"m3x_merkleize_chunks_dev rc={rc}")
+    return out.raw
+
+
+def basic_list_root_dev(dev_ptr, n_elems: int, elem_size: int,
+                        limit_elems: int, ctx=None) -> bytes:
+    limit_chunks = max((limit_elems * elem_size + 31) // 32, 1)
+    n_chunks = (n_elems * elem_size + 31) // 32
+    return merkleize_chunks_dev(dev_ptr, n_chunks, _ceil_log2(limit_chunks),
+                                n_elems, ctx=ctx)
+
+
+def basic_vector_root_dev(dev_ptr, n_elems: int, elem_size: int, ctx=None) -> bytes:
+    limit_chunks = max((n_elems * elem_size + 31) // 32, 1)
+    n_chunks = (n_elems * elem_size + 31) // 32
+    return merkleize_chunks_dev(dev_ptr, n_chunks, _ceil_log2(limit_chunks),
+                                -1, ctx=ctx)
+
+
+def root_vector_root_dev(dev_ptr, n: int, ctx=None) -> bytes:
+    return merkleize_chunks_dev(dev_ptr, n, _ceil_log2(max(n, 1)), -1, ctx=ctx)
+
+
+def validator_registry_root_dev(dev_ptr, n: int, ctx=None) -> bytes:
+    import ctypes as ct
+
+    ctx = ctx or _native.default_ctx()
+    out = ct.create_string_buffer(32)
+    rc = ctx._lib.m3x_merkleize_validators_dev(ctx.handle, dev_ptr, n, out)
+    if rc != 0:
+        raise RuntimeError(f"m3x_merkleize_validators_dev rc={rc}")
+    return out.raw
+
+
 def root_vector_root(roots: bytes, n: int, ctx=None) -> bytes:
     return merkleize_chunks(roots, n, _ceil_log2(max(n, 1)), -1, ctx=ctx)
 

@@ -95,3 +95,97 @@ def synthetic_validator_ssz(i: int) -> bytes:
     ee = (2**64 - 1).to_bytes(8, "little")
     we = (2**64 - 1).to_bytes(8, "little")
     return pk + wc + eff + slashed + aee + ae + ee + we
+
+
+# ---- full Deneb BeaconState root (mirror of lighthouse_amd.beacon_state,
+# computed with hashlib; used to pin the GPU composition) ----
+
+def _c(b):
+    return b + b"\x00" * (32 - len(b))
+
+
+def _u64le(v):
+    return int(v).to_bytes(8, "little")
+
+
+def _container(chunks):
+    d = 0
+    while (1 << d) < len(chunks):
+        d += 1
+    return merkleize(chunks, d)
+
+
+def _list_of_roots(roots, n, limit):
+    return mix_in_length(merkleize(roots, ceil_log2(max(limit, 1))), n)
+
+
+def beacon_state_root_ref(st):
+    f = []
+    f.append(_c(_u64le(st["genesis_time"])))
+    f.append(st["genesis_validators_root"])
+    f.append(_c(_u64le(st["slot"])))
+    pv, cv, ep = st["fork"]
+    f.append(_container([_c(pv), _c(cv), _c(_u64le(ep))]))
+    sl, pi, pr, sr, br = st["latest_block_header"]
+    f.append(_container([_c(_u64le(sl)), _c(_u64le(pi)), pr, sr, br]))
+    f.append(merkleize(pack_bytes(st["block_roots"]), 13))
+    f.append(merkleize(pack_bytes(st["state_roots"]), 13))
+    f.append(_list_of_roots(pack_bytes(st["historical_roots"]),
+                            len(st["historical_roots"]) // 32, 1 << 24))
+
+    def eth1_root(e):
+        dr, dc, bh = e
+        return _container([dr, _c(_u64le(dc)), bh])
+
+    f.append(eth1_root(st["eth1_data"]))
+    f.append(_list_of_roots([eth1_root(v) for v in st["eth1_data_votes"]],
+                            len(st["eth1_data_votes"]), 2048))
+    f.append(_c(_u64le(st["eth1_deposit_index"])))
+    f.append(validator_registry_root(st["validators_ssz"], st["n_validators"]))
+    f.append(basic_list_root(st["balances"], st["n_validators"], 8, 1 << 40))
+    f.append(merkleize(pack_bytes(st["randao_mixes"]), 16))
+    f.append(basic_vector_root(st["slashings"], 8192, 8))
+    f.append(basic_list_root(st["previous_epoch_participation"],
+                             st["n_validators"], 1, 1 << 40))
+    f.append(basic_list_root(st["current_epoch_participation"],
+                             st["n_validators"], 1, 1 << 40))
+    f.append(_c(st["justification_bits"]))
+
+    def ckpt(c):
+        e, r = c
+        return _container([_c(_u64le(e)), r])
+
+    f.append(ckpt(st["previous_justified_checkpoint"]))
+    f.append(ckpt(st["current_justified_checkpoint"]))
+    f.append(ckpt(st["finalized_checkpoint"]))
+    f.append(basic_list_root(st["inactivity_scores"], st["n_validators"], 8,
+                             1 << 40))
+
+    def sync_root(sc):
+        pks, agg = sc
+        leaves = [merkleize([_c(pk[:32]), _c(pk[32:48])], 1) for pk in pks]
+        return _container([merkleize(leaves, 9),
+                           merkleize([_c(agg[:32]), _c(agg[32:48])], 1)])
+
+    f.append(sync_root(st["current_sync_committee"]))
+    f.append(sync_root(st["next_sync_committee"]))
+    h = st["latest_execution_payload_header"]
+    ph = [
+        h["parent_hash"], _c(h["fee_recipient"]), h["state_root"],
+        h["receipts_root"], merkleize(pack_bytes(h["logs_bloom"]), 3),
+        h["prev_randao"], _c(_u64le(h["block_number"])),
+        _c(_u64le(h["gas_limit"])), _c(_u64le(h["gas_used"])),
+        _c(_u64le(h["timestamp"])),
+        basic_list_root(h["extra_data"], len(h["extra_data"]), 1, 32),
+        h["base_fee_per_gas"], h["block_hash"], h["transactions_root"],
+        h["withdrawals_root"], _c(_u64le(h["blob_gas_used"])),
+        _c(_u64le(h["excess_blob_gas"])),
+    ]
+    f.append(_container(ph))
+    f.append(_c(_u64le(st["next_withdrawal_index"])))
+    f.append(_c(_u64le(st["next_withdrawal_validator_index"])))
+    f.append(_list_of_roots([_container([a, b])
+                             for a, b in st["historical_summaries"]],
+                            len(st["historical_summaries"]), 1 << 24))
+    assert len(f) == 28
+    return _container(f)
