@@ -378,9 +378,8 @@ def main():
             }
         leaves_ms, leaves_n = ktimes.get("leaves", (0.0, 0))
         reduce_ms, reduce_n = ktimes.get("reduce", (0.0, 0))
-        sha_hps = (
-            REGISTRY_NODE_HASHES / merkle_only if merkle_only > 0 else None
-        )
+        full_nodes = bs.node_hash_count(N_VALIDATORS)
+        sha_hps = full_nodes / merkle_only if merkle_only > 0 else None
         line = {
             "metric": "bls_sig_sets_verified_per_sec",
             "value": value,
