@@ -96,6 +96,14 @@ int32_t m3x_validator_subtree_root_dev(m3x_ctx *ctx, const void *ssz_dev,
                                        uint64_t n, uint32_t depth,
                                        uint8_t out_root[32]);
 
+/* Batched small-container merkleize: element i owns chunks
+ * [offsets[i], offsets[i+1]) (<=32 chunks each); out_roots[i] = its
+ * hash_tree_root (depth = ceil_log2(count)). One launch for the
+ * BeaconState's many tiny containers. */
+int32_t m3x_merkleize_batch(m3x_ctx *ctx, const uint8_t *chunks,
+                            const uint32_t *offsets, uint64_t n_elems,
+                            uint8_t *out_roots);
+
 /* ---- BLS12-381 batched signature-set verification (hot path #1) ---- */
 
 /* key_validate a batch of compressed pubkeys: decompress + infinity reject +

@@ -84,6 +84,16 @@ def _bind(lib):
         ),
         ("m3x_timing_enable", [ctypes.c_void_p, ctypes.c_int32]),
         (
+            "m3x_merkleize_batch",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.POINTER(ctypes.c_uint32),
+                ctypes.c_uint64,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
             "m3x_kernel_ms",
             [
                 ctypes.c_void_p,

@@ -136,123 +136,119 @@ def state_root(st: dict, ctx=None, dev=None, registry_root=None) -> bytes:
     """hash_tree_root of the synthetic Deneb state — every hash on GPU.
     With `dev` (from upload_fields) the big fields are read from HBM;
     with `registry_root` the validators field root is taken as given
-    (multi-GPU sharded path computes it collectively)."""
-    C = lambda b: _chunk(b)
-    f = []  # 28 field roots, SSZ order
+    (multi-GPU sharded path computes it collectively). The many tiny
+    containers go through ONE m3x_merkleize_batch launch; the full
+    sync-committee pubkey vectors flatten into single merkleize calls
+    (valid because the vectors are full: element roots at depth 1 of the
+    combined chunk tree equal the per-element roots)."""
+    from . import _native
 
-    f.append(C(_u64(st["genesis_time"])))
-    f.append(st["genesis_validators_root"])
-    f.append(C(_u64(st["slot"])))
+    ctx = ctx or _native.default_ctx()
+    C = _chunk
+    u = _u64
+
+    # ---- batch 1: every independent small container ----
+    groups = []
+
+    def grp(*chunks):
+        groups.append(b"".join(chunks))
+        return len(groups) - 1
+
     pv, cv, ep = st["fork"]
-    f.append(_container_root([C(pv), C(cv), C(_u64(ep))], ctx))
+    gi_fork = grp(C(pv), C(cv), C(u(ep)))
     sl, pi, pr, sr, br = st["latest_block_header"]
-    f.append(_container_root([C(_u64(sl)), C(_u64(pi)), pr, sr, br], ctx))
-    if dev:
-        f.append(th.root_vector_root_dev(dev["block_roots"],
-                                         SLOTS_PER_HISTORICAL_ROOT, ctx))
-        f.append(th.root_vector_root_dev(dev["state_roots"],
-                                         SLOTS_PER_HISTORICAL_ROOT, ctx))
-    else:
-        f.append(th.root_vector_root(st["block_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx))
-        f.append(th.root_vector_root(st["state_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx))
-    f.append(th.root_list_root(st["historical_roots"],
-                               len(st["historical_roots"]) // 32,
-                               HISTORICAL_ROOTS_LIMIT, ctx))
+    gi_hdr = grp(C(u(sl)), C(u(pi)), pr, sr, br)
+    dr, dc, bh = st["eth1_data"]
+    gi_eth1 = grp(dr, C(u(dc)), bh)
+    gi_votes = [grp(v[0], C(u(v[1])), v[2]) for v in st["eth1_data_votes"]]
+    gi_pjc = grp(C(u(st["previous_justified_checkpoint"][0])),
+                 st["previous_justified_checkpoint"][1])
+    gi_cjc = grp(C(u(st["current_justified_checkpoint"][0])),
+                 st["current_justified_checkpoint"][1])
+    gi_fc = grp(C(u(st["finalized_checkpoint"][0])),
+                st["finalized_checkpoint"][1])
+    h = st["latest_execution_payload_header"]
+    # payload needs two sub-roots first (logs_bloom, extra_data) — both are
+    # fixed 8-chunk / 1-chunk basic vectors: fold them in the same batch
+    gi_bloom = grp(*[h["logs_bloom"][32 * i : 32 * (i + 1)] for i in range(8)])
+    gi_agg1 = grp(C(st["current_sync_committee"][1][:32]),
+                  C(st["current_sync_committee"][1][32:48]))
+    gi_agg2 = grp(C(st["next_sync_committee"][1][:32]),
+                  C(st["next_sync_committee"][1][32:48]))
+    gi_hs = [grp(a, b) for a, b in st["historical_summaries"]]
+    roots = th.merkleize_batch(groups, ctx=ctx)
 
-    def eth1_root(e):
-        dr, dc, bh = e
-        return _container_root([dr, C(_u64(dc)), bh], ctx)
-
-    f.append(eth1_root(st["eth1_data"]))
-    votes = b"".join(eth1_root(v) for v in st["eth1_data_votes"])
-    f.append(th.root_list_root(votes, len(st["eth1_data_votes"]),
-                               SLOTS_PER_ETH1_VOTING, ctx))
-    f.append(C(_u64(st["eth1_deposit_index"])))
+    # ---- big/independent fields ----
+    f6 = (th.root_vector_root_dev(dev["block_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx)
+          if dev else th.root_vector_root(st["block_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx))
+    f7 = (th.root_vector_root_dev(dev["state_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx)
+          if dev else th.root_vector_root(st["state_roots"], SLOTS_PER_HISTORICAL_ROOT, ctx))
+    f8 = th.root_list_root(st["historical_roots"],
+                           len(st["historical_roots"]) // 32,
+                           HISTORICAL_ROOTS_LIMIT, ctx)
+    votes_roots = b"".join(roots[i] for i in gi_votes)
+    f10 = th.root_list_root(votes_roots, len(gi_votes), SLOTS_PER_ETH1_VOTING, ctx)
     n = st["n_validators"]
     if registry_root is not None:
-        f.append(registry_root)
+        f12 = registry_root
     elif dev:
-        f.append(th.validator_registry_root_dev(dev["validators_ssz"], n, ctx))
+        f12 = th.validator_registry_root_dev(dev["validators_ssz"], n, ctx)
     else:
-        f.append(th.validator_registry_root(st["validators_ssz"], n, ctx))
+        f12 = th.validator_registry_root(st["validators_ssz"], n, ctx)
     if dev:
-        f.append(th.basic_list_root_dev(dev["balances"], n, 8, REGISTRY_LIMIT, ctx))
-        f.append(th.root_vector_root_dev(dev["randao_mixes"],
-                                         EPOCHS_PER_HISTORICAL_VECTOR, ctx))
-        f.append(th.basic_vector_root_dev(dev["slashings"], EPOCHS_PER_SLASHINGS, 8, ctx))
-        f.append(th.basic_list_root_dev(dev["previous_epoch_participation"],
-                                        n, 1, REGISTRY_LIMIT, ctx))
-        f.append(th.basic_list_root_dev(dev["current_epoch_participation"],
-                                        n, 1, REGISTRY_LIMIT, ctx))
+        f13 = th.basic_list_root_dev(dev["balances"], n, 8, REGISTRY_LIMIT, ctx)
+        f14 = th.root_vector_root_dev(dev["randao_mixes"], EPOCHS_PER_HISTORICAL_VECTOR, ctx)
+        f15 = th.basic_vector_root_dev(dev["slashings"], EPOCHS_PER_SLASHINGS, 8, ctx)
+        f16 = th.basic_list_root_dev(dev["previous_epoch_participation"], n, 1, REGISTRY_LIMIT, ctx)
+        f17 = th.basic_list_root_dev(dev["current_epoch_participation"], n, 1, REGISTRY_LIMIT, ctx)
+        f22 = th.basic_list_root_dev(dev["inactivity_scores"], n, 8, REGISTRY_LIMIT, ctx)
     else:
-        f.append(th.basic_list_root(st["balances"], n, 8, REGISTRY_LIMIT, ctx))
-        f.append(th.root_vector_root(st["randao_mixes"],
-                                     EPOCHS_PER_HISTORICAL_VECTOR, ctx))
-        f.append(th.basic_vector_root(st["slashings"], EPOCHS_PER_SLASHINGS, 8, ctx))
-        f.append(th.basic_list_root(st["previous_epoch_participation"],
-                                    n, 1, REGISTRY_LIMIT, ctx))
-        f.append(th.basic_list_root(st["current_epoch_participation"],
-                                    n, 1, REGISTRY_LIMIT, ctx))
-    f.append(C(st["justification_bits"]))  # BitVector[4]: one chunk
+        f13 = th.basic_list_root(st["balances"], n, 8, REGISTRY_LIMIT, ctx)
+        f14 = th.root_vector_root(st["randao_mixes"], EPOCHS_PER_HISTORICAL_VECTOR, ctx)
+        f15 = th.basic_vector_root(st["slashings"], EPOCHS_PER_SLASHINGS, 8, ctx)
+        f16 = th.basic_list_root(st["previous_epoch_participation"], n, 1, REGISTRY_LIMIT, ctx)
+        f17 = th.basic_list_root(st["current_epoch_participation"], n, 1, REGISTRY_LIMIT, ctx)
+        f22 = th.basic_list_root(st["inactivity_scores"], n, 8, REGISTRY_LIMIT, ctx)
 
-    def ckpt_root(c):
-        e, r = c
-        return _container_root([C(_u64(e)), r], ctx)
+    # sync-committee pubkeys: FULL Vector[Bytes48, 512] -> flattened
+    # 1024-chunk tree at depth 10 (single call)
+    def pks_root(pks):
+        flat = b"".join(C(pk[:32]) + C(pk[32:48]) for pk in pks)
+        return th.merkleize_chunks(flat, 1024, 10, -1, ctx=ctx)
 
-    f.append(ckpt_root(st["previous_justified_checkpoint"]))
-    f.append(ckpt_root(st["current_justified_checkpoint"]))
-    f.append(ckpt_root(st["finalized_checkpoint"]))
-    if dev:
-        f.append(th.basic_list_root_dev(dev["inactivity_scores"], n, 8,
-                                        REGISTRY_LIMIT, ctx))
-    else:
-        f.append(th.basic_list_root(st["inactivity_scores"], n, 8,
-                                    REGISTRY_LIMIT, ctx))
+    pks1 = pks_root(st["current_sync_committee"][0])
+    pks2 = pks_root(st["next_sync_committee"][0])
+    hs_roots = b"".join(roots[i] for i in gi_hs)
+    f28 = th.root_list_root(hs_roots, len(gi_hs), HISTORICAL_ROOTS_LIMIT, ctx)
 
-    def sync_root(sc):
-        pks, agg = sc
-        leaves = b"".join(
-            th.merkleize_chunks(_chunk(pk[:32]) + _chunk(pk[32:48]), 2, 1, -1,
-                                ctx=ctx)
-            for pk in pks
-        )
-        pks_root = th.merkleize_chunks(leaves, len(pks), 9, -1, ctx=ctx)
-        agg_root = th.merkleize_chunks(
-            _chunk(agg[:32]) + _chunk(agg[32:48]), 2, 1, -1, ctx=ctx
-        )
-        return _container_root([pks_root, agg_root], ctx)
-
-    f.append(sync_root(st["current_sync_committee"]))
-    f.append(sync_root(st["next_sync_committee"]))
-
-    h = st["latest_execution_payload_header"]
-    ph = [
-        h["parent_hash"],
-        C(h["fee_recipient"]),
-        h["state_root"],
-        h["receipts_root"],
-        th.basic_vector_root(h["logs_bloom"], 256, 1, ctx),
-        h["prev_randao"],
-        C(_u64(h["block_number"])),
-        C(_u64(h["gas_limit"])),
-        C(_u64(h["gas_used"])),
-        C(_u64(h["timestamp"])),
-        th.basic_list_root(h["extra_data"], len(h["extra_data"]), 1, 32, ctx),
-        h["base_fee_per_gas"],
-        h["block_hash"],
-        h["transactions_root"],
-        h["withdrawals_root"],
-        C(_u64(h["blob_gas_used"])),
-        C(_u64(h["excess_blob_gas"])),
+    # ---- batch 2: containers that depend on batch-1/sub roots ----
+    groups2 = [
+        pks1 + roots[gi_agg1],  # current_sync_committee
+        pks2 + roots[gi_agg2],  # next_sync_committee
+        # payload header (17 field chunks)
+        b"".join([
+            h["parent_hash"], C(h["fee_recipient"]), h["state_root"],
+            h["receipts_root"], roots[gi_bloom], h["prev_randao"],
+            C(u(h["block_number"])), C(u(h["gas_limit"])),
+            C(u(h["gas_used"])), C(u(h["timestamp"])),
+            th.basic_list_root(h["extra_data"], len(h["extra_data"]), 1, 32, ctx),
+            h["base_fee_per_gas"], h["block_hash"], h["transactions_root"],
+            h["withdrawals_root"], C(u(h["blob_gas_used"])),
+            C(u(h["excess_blob_gas"])),
+        ]),
     ]
-    f.append(_container_root(ph, ctx))
-    f.append(C(_u64(st["next_withdrawal_index"])))
-    f.append(C(_u64(st["next_withdrawal_validator_index"])))
-    hs = b"".join(_container_root([a, b], ctx) for a, b in st["historical_summaries"])
-    f.append(th.root_list_root(hs, len(st["historical_summaries"]),
-                               HISTORICAL_ROOTS_LIMIT, ctx))
+    r2 = th.merkleize_batch(groups2, ctx=ctx)
 
-    assert len(f) == 28, len(f)
+    f = [
+        C(u(st["genesis_time"])), st["genesis_validators_root"],
+        C(u(st["slot"])), roots[gi_fork], roots[gi_hdr], f6, f7, f8,
+        roots[gi_eth1], f10, C(u(st["eth1_deposit_index"])), f12, f13, f14,
+        f15, f16, f17, C(st["justification_bits"]), roots[gi_pjc],
+        roots[gi_cjc], roots[gi_fc], f22, r2[0], r2[1], r2[2],
+        C(u(st["next_withdrawal_index"])),
+        C(u(st["next_withdrawal_validator_index"])), f28,
+    ]
+    assert len(f) == 28
     return _container_root(f, ctx)
 
 

@@ -292,7 +292,7 @@ extern "C" {
 int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp, uint64_t n,
                               uint8_t *uncomp, int32_t *status) {
   if (!ctx || n == 0) return M3X_ERR_ARG;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
   uint8_t *comp_d, *unc_d;
   int32_t *st_d;
@@ -318,7 +318,7 @@ int32_t m3x_bls_verify_sets_dev(m3x_ctx *ctx, const void *msgs_dev,
                                 const void *rands_dev, uint64_t n) {
   if (!ctx) return M3X_ERR_ARG;
   if (n == 0) return 0; // blst.rs:42-44 (host normally short-circuits)
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
   int32_t verdict = 0;
   int rc = run_verify(ctx, msgs_dev, sigs_dev, pks_dev, pk_offsets_dev,

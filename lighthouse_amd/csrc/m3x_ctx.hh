@@ -27,6 +27,10 @@ struct m3x_ctx {
   uint64_t scratch_a_bytes = 0;
   uint8_t *scratch_b = nullptr;
   uint64_t scratch_b_bytes = 0;
+  // persistent pool for small inputs + 32B root slots (avoids per-call
+  // hipMalloc/hipFree on the many tiny container merkleizations)
+  uint8_t *small_pool = nullptr; // layout: [256 KiB input][64B root]
+  static constexpr uint64_t SMALL_POOL_IN = 256 * 1024;
   // per-kernel cumulative time (ms) + launch counts since last reset,
   // measured with hipEvents on `stream`
   hipEvent_t ev_a[M3X_K_COUNT] = {};
@@ -34,7 +38,7 @@ struct m3x_ctx {
   double kernel_ms[M3X_K_COUNT] = {};
   uint64_t kernel_launches[M3X_K_COUNT] = {};
   bool timing = false;
-  std::mutex mu; // one context serializes its own calls
+  std::recursive_mutex mu; // one context serializes its own calls
 };
 
 namespace m3x {

@@ -245,7 +245,7 @@ int32_t m3x_abi_version(void) { return 1; }
 
 int32_t m3x_timing_enable(m3x_ctx *ctx, int32_t on) {
   if (!ctx) return M3X_ERR_ARG;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   ctx->timing = on != 0;
   for (int i = 0; i < M3X_K_COUNT; i++) {
     ctx->kernel_ms[i] = 0;
@@ -268,14 +268,11 @@ int32_t m3x_finalize_root(m3x_ctx *ctx, const uint8_t node[32],
                           uint32_t from_level, uint32_t to_depth,
                           int64_t mix_len, uint8_t out_root[32]) {
   if (!ctx) return M3X_ERR_ARG;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
-  uint8_t *tmp;
-  M3X_HIP_CHECK(hipMalloc(&tmp, 64));
-  if (hipMemcpy(tmp, node, 32, hipMemcpyHostToDevice) != hipSuccess) {
-    (void)hipFree(tmp);
+  uint8_t *tmp = ctx->small_pool;
+  if (hipMemcpy(tmp, node, 32, hipMemcpyHostToDevice) != hipSuccess)
     return M3X_ERR_HIP;
-  }
   hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream, tmp,
                      from_level, to_depth, mix_len, ctx->zeros_dev, tmp + 32);
   int32_t r2 = M3X_OK;
@@ -283,7 +280,6 @@ int32_t m3x_finalize_root(m3x_ctx *ctx, const uint8_t node[32],
                      ctx->stream) != hipSuccess)
     r2 = M3X_ERR_HIP;
   if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
-  (void)hipFree(tmp);
   return r2;
 }
 
@@ -302,6 +298,11 @@ int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
     delete ctx;
     return M3X_ERR_NOMEM;
   }
+  if (hipMalloc(&ctx->small_pool, m3x_ctx::SMALL_POOL_IN + 64) !=
+      hipSuccess) {
+    delete ctx;
+    return M3X_ERR_NOMEM;
+  }
   hipLaunchKernelGGL(k_zero_ladder, dim3(1), dim3(64), 0, ctx->stream,
                      ctx->zeros_dev);
   if (hipStreamSynchronize(ctx->stream) != hipSuccess) {
@@ -315,6 +316,7 @@ int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
 void m3x_ctx_destroy(m3x_ctx *ctx) {
   if (!ctx) return;
   if (ctx->zeros_dev) (void)hipFree(ctx->zeros_dev);
+  if (ctx->small_pool) (void)hipFree(ctx->small_pool);
   if (ctx->scratch_a) (void)hipFree(ctx->scratch_a);
   if (ctx->scratch_b) (void)hipFree(ctx->scratch_b);
   if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
@@ -394,10 +396,9 @@ int32_t m3x_merkleize_chunks_dev(m3x_ctx *ctx, const void *chunks_dev,
                                  int64_t mix_len, uint8_t out_root[32]) {
   if (!ctx) return M3X_ERR_ARG;
   if (n_chunks > (1ull << depth) && depth < 63) return M3X_ERR_ARG;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
-  uint8_t *root_dev;
-  M3X_HIP_CHECK(hipMalloc(&root_dev, 64));
+  uint8_t *root_dev = ctx->small_pool + m3x_ctx::SMALL_POOL_IN;
   int32_t rc;
   if (n_chunks == 0) {
     // Z[depth] then optional mix
@@ -420,7 +421,6 @@ int32_t m3x_merkleize_chunks_dev(m3x_ctx *ctx, const void *chunks_dev,
                      ctx->stream) != hipSuccess)
     r2 = M3X_ERR_HIP;
   if (hipStreamSynchronize(ctx->stream) != hipSuccess) r2 = M3X_ERR_HIP;
-  (void)hipFree(root_dev);
   return r2;
 }
 
@@ -428,19 +428,26 @@ int32_t m3x_merkleize_chunks(m3x_ctx *ctx, const uint8_t *chunks,
                              uint64_t n_chunks, uint32_t depth,
                              int64_t mix_len, uint8_t out_root[32]) {
   if (!ctx) return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
   void *dev = nullptr;
+  bool pooled = false;
   uint64_t bytes = n_chunks * 32;
   if (bytes) {
-    M3X_HIP_CHECK(hipMalloc(&dev, bytes));
+    if (bytes <= m3x_ctx::SMALL_POOL_IN) {
+      dev = ctx->small_pool;
+      pooled = true;
+    } else {
+      M3X_HIP_CHECK(hipMalloc(&dev, bytes));
+    }
     if (hipMemcpy(dev, chunks, bytes, hipMemcpyHostToDevice) != hipSuccess) {
-      (void)hipFree(dev);
+      if (!pooled) (void)hipFree(dev);
       return M3X_ERR_HIP;
     }
   }
   int32_t rc =
       m3x_merkleize_chunks_dev(ctx, dev, n_chunks, depth, mix_len, out_root);
-  if (dev) (void)hipFree(dev);
+  if (dev && !pooled) (void)hipFree(dev);
   return rc;
 }
 
@@ -448,7 +455,7 @@ int32_t m3x_validator_subtree_root_dev(m3x_ctx *ctx, const void *ssz_dev,
                                        uint64_t n, uint32_t depth,
                                        uint8_t out_root[32]) {
   if (!ctx || (n > (1ull << depth) && depth < 63)) return M3X_ERR_ARG;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
   // leaves
   uint64_t n_pad = n ? n : 1;
@@ -489,7 +496,7 @@ int32_t m3x_merkleize_validators_dev(m3x_ctx *ctx, const void *ssz_dev,
   uint8_t sub[32];
   int32_t rc = m3x_validator_subtree_root_dev(ctx, ssz_dev, n, 40, sub);
   if (rc != M3X_OK) return rc;
-  std::lock_guard<std::mutex> lk(ctx->mu);
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   uint8_t *tmp;
   M3X_HIP_CHECK(hipMalloc(&tmp, 64));
   if (hipMemcpy(tmp, sub, 32, hipMemcpyHostToDevice) != hipSuccess) {
@@ -527,3 +534,92 @@ int32_t m3x_merkleize_validators(m3x_ctx *ctx, const uint8_t *ssz, uint64_t n,
 }
 
 } // extern "C"
+
+// Batched small-container merkleize: one thread folds one element's chunk
+// group (<=32 chunks, depth = ceil_log2(count)) — the BeaconState's many
+// tiny containers (Fork, headers, Eth1Data votes, checkpoints,
+// HistoricalSummary, payload header) in ONE launch.
+__global__ void k_merkleize_batch(const uint8_t *__restrict__ chunks,
+                                  const uint32_t *__restrict__ offs,
+                                  uint64_t n, const uint8_t *__restrict__ zeros,
+                                  uint8_t *__restrict__ out) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint32_t c0 = offs[i], c1 = offs[i + 1];
+  uint32_t cnt = c1 - c0;
+  uint32_t node[32][8];
+  for (uint32_t j = 0; j < cnt && j < 32; j++) {
+    const uint32_t *src =
+        reinterpret_cast<const uint32_t *>(chunks + 32ull * (c0 + j));
+    for (int q = 0; q < 8; q++) node[j][q] = __builtin_bswap32(src[q]);
+  }
+  uint32_t depth = 0;
+  while ((1u << depth) < cnt) depth++;
+  uint32_t m = cnt, level = 0;
+  while (level < depth) {
+    uint32_t nx = (m + 1) / 2;
+    for (uint32_t q = 0; q < nx; q++) {
+      uint32_t zw[8];
+      const uint32_t *r;
+      if (2 * q + 1 < m) {
+        r = node[2 * q + 1];
+      } else {
+        for (int b = 0; b < 8; b++)
+          zw[b] = be_load_u8x4(zeros + 32 * level + 4 * b);
+        r = zw;
+      }
+      uint32_t o[8];
+      sha256_node(node[2 * q], r, o);
+      for (int b = 0; b < 8; b++) node[q][b] = o[b];
+    }
+    m = nx;
+    level++;
+  }
+  uint32_t *dst = reinterpret_cast<uint32_t *>(out + 32 * i);
+  for (int b = 0; b < 8; b++) dst[b] = __builtin_bswap32(node[0][b]);
+}
+
+extern "C" int32_t m3x_merkleize_batch(m3x_ctx *ctx, const uint8_t *chunks,
+                                       const uint32_t *offsets,
+                                       uint64_t n_elems,
+                                       uint8_t *out_roots) {
+  if (!ctx || n_elems == 0) return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint64_t in_bytes = (uint64_t)offsets[n_elems] * 32;
+  uint64_t off_bytes = (n_elems + 1) * 4;
+  uint64_t out_bytes = n_elems * 32;
+  uint64_t need = in_bytes + off_bytes + out_bytes + 256;
+  uint8_t *base;
+  bool pooled = need <= m3x_ctx::SMALL_POOL_IN;
+  if (pooled) {
+    base = ctx->small_pool;
+  } else {
+    M3X_HIP_CHECK(hipMalloc(&base, need));
+  }
+  uint8_t *in_d = base;
+  uint8_t *off_d = base + in_bytes;
+  uint8_t *out_d = base + in_bytes + ((off_bytes + 31) & ~31ull);
+  int32_t rc = M3X_OK;
+  do {
+    if (hipMemcpyAsync(in_d, chunks, in_bytes, hipMemcpyHostToDevice,
+                       ctx->stream) != hipSuccess ||
+        hipMemcpyAsync(off_d, offsets, off_bytes, hipMemcpyHostToDevice,
+                       ctx->stream) != hipSuccess) {
+      rc = M3X_ERR_HIP;
+      break;
+    }
+    uint32_t blocks = (uint32_t)((n_elems + 63) / 64);
+    hipLaunchKernelGGL(k_merkleize_batch, dim3(blocks), dim3(64), 0,
+                       ctx->stream, in_d, (const uint32_t *)off_d, n_elems,
+                       ctx->zeros_dev, out_d);
+    if (hipMemcpyAsync(out_roots, out_d, out_bytes, hipMemcpyDeviceToHost,
+                       ctx->stream) != hipSuccess) {
+      rc = M3X_ERR_HIP;
+      break;
+    }
+    if (hipStreamSynchronize(ctx->stream) != hipSuccess) rc = M3X_ERR_HIP;
+  } while (0);
+  if (!pooled) (void)hipFree(base);
+  return rc;
+}

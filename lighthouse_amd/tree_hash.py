@@ -107,3 +107,25 @@ def root_vector_root(roots: bytes, n: int, ctx=None) -> bytes:
 
 def root_list_root(roots: bytes, n: int, limit: int, ctx=None) -> bytes:
     return merkleize_chunks(roots, n, _ceil_log2(max(limit, 1)), n, ctx=ctx)
+
+
+def merkleize_batch(chunk_groups, ctx=None):
+    """One GPU launch for many tiny containers: chunk_groups is a list of
+    bytes (each a concatenation of 32B chunks, <=32 chunks per group);
+    returns the list of hash_tree_roots (depth = ceil_log2(count))."""
+    import ctypes as ct
+
+    ctx = ctx or _native.default_ctx()
+    offsets = [0]
+    blob = bytearray()
+    for g in chunk_groups:
+        assert len(g) % 32 == 0 and len(g) <= 32 * 32
+        blob += g
+        offsets.append(offsets[-1] + len(g) // 32)
+    n = len(chunk_groups)
+    off = (ct.c_uint32 * (n + 1))(*offsets)
+    out = ct.create_string_buffer(32 * n)
+    rc = ctx._lib.m3x_merkleize_batch(ctx.handle, bytes(blob), off, n, out)
+    if rc != 0:
+        raise RuntimeError(f"m3x_merkleize_batch rc={rc}")
+    return [out.raw[32 * i : 32 * (i + 1)] for i in range(n)]

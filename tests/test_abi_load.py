@@ -18,6 +18,10 @@ EXPECTED = [
     "m3x_merkleize_validators_dev",
     "m3x_merkleize_chunks",
     "m3x_merkleize_chunks_dev",
+    "m3x_merkleize_batch",
+    "m3x_finalize_root",
+    "m3x_timing_enable",
+    "m3x_kernel_ms",
     "m3x_validator_subtree_root_dev",
     "m3x_bls_pk_decompress",
     "m3x_bls_verify_sets",
