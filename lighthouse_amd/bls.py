@@ -148,6 +148,47 @@ def _draw_rands(n):
     return out
 
 
+def verify(signature: "Signature", pubkey: PublicKey, msg: bytes,
+           ctx=None) -> bool:
+    """Single-set verify (TSignature::verify, blst.rs:196-200): the batch
+    equation with r=1 degenerates to e(pk, H(m)) == e(g1, sigma)."""
+    if signature.is_empty():
+        return False
+    s = SignatureSet(signature, [pubkey], msg)
+    return verify_signature_sets([s], ctx=ctx, _rands=[1])
+
+
+def fast_aggregate_verify(signature: "Signature", msg: bytes, pubkeys,
+                          ctx=None) -> bool:
+    """TAggregateSignature::fast_aggregate_verify (blst.rs:250-261;
+    generic_aggregate_signature.rs:187-196): one message, k pubkeys."""
+    pubkeys = list(pubkeys)
+    if not pubkeys:
+        return False
+    if signature.is_empty():
+        return False
+    s = SignatureSet(signature, pubkeys, msg)
+    return verify_signature_sets([s], ctx=ctx, _rands=[1])
+
+
+def eth_fast_aggregate_verify(signature: "Signature", msg: bytes, pubkeys,
+                              ctx=None) -> bool:
+    """generic_aggregate_signature.rs:198-210: accepts the
+    G2_POINT_AT_INFINITY signature when pubkeys is empty (sync-aggregate
+    rule)."""
+    pubkeys = list(pubkeys)
+    if not pubkeys and signature.serialize() == INFINITY_SIGNATURE:
+        return True
+    return fast_aggregate_verify(signature, msg, pubkeys, ctx=ctx)
+
+
+# aggregate_verify (distinct messages under ONE aggregate signature,
+# blst.rs:263-274) exists in the reference "only for EF tests, presently
+# not used in production" (generic_aggregate_signature.rs:44-47); it is not
+# expressible as independent signature sets and is served by the oracle in
+# tests until a dedicated kernel is warranted.
+
+
 def verify_signature_sets(sets, ctx=None, _rands=None) -> bool:
     """bls::verify_signature_sets (blst.rs:37-119). `_rands` exists for
     deterministic tests only."""

@@ -167,3 +167,44 @@ def test_aggregate_sets_k_gt_1(ctx, oracle):
     )
     sets[1].signing_keys[0] = other
     assert bls.verify_signature_sets(sets, ctx=ctx) is False
+
+
+def test_single_and_fast_aggregate_verify(ctx, oracle):
+    """single-set paths (blst.rs:196-200, :250-261) vs the oracle."""
+    from lighthouse_amd import bls
+
+    it = FIXTURES["signatures"][0]
+    pk = bls.PublicKey.from_uncompressed(
+        bytes.fromhex(FIXTURES["interop"][0]["pk_uncompressed_hex"])
+    )
+    msg = bytes.fromhex(it["msg_hex"])
+    sig = bls.Signature.from_compressed(bytes.fromhex(it["sig_compressed_hex"]))
+    assert bls.verify(sig, pk, msg, ctx=ctx) is True
+    assert bls.verify(sig, pk, b"\x42" * 32, ctx=ctx) is False
+    assert (
+        oracle.m3x_oracle_bls_verify(
+            bytes.fromhex(FIXTURES["interop"][0]["pk_uncompressed_hex"]),
+            msg,
+            bytes.fromhex(it["sig_compressed_hex"]),
+        )
+        == 1
+    )
+    # fast_aggregate_verify on an aggregate case from the fixtures
+    case = next(
+        c for c in FIXTURES["batch_cases"] if c["name"] == "three_valid_mixed_k"
+    )
+    s = case["sets"][1]
+    agg_sig = bls.Signature.from_compressed(
+        bytes.fromhex(s["sig_compressed_hex"])
+    )
+    pks = [
+        bls.PublicKey.from_uncompressed(bytes.fromhex(p))
+        for p in s["pks_uncompressed_hex"]
+    ]
+    assert bls.fast_aggregate_verify(agg_sig, bytes.fromhex(s["msg_hex"]), pks, ctx=ctx)
+    assert not bls.fast_aggregate_verify(agg_sig, b"\x13" * 32, pks, ctx=ctx)
+    assert not bls.fast_aggregate_verify(agg_sig, bytes.fromhex(s["msg_hex"]), [], ctx=ctx)
+    # eth variant: infinity sig + empty pubkeys is valid (sync aggregate)
+    assert bls.eth_fast_aggregate_verify(
+        bls.Signature.infinity(), b"\x00" * 32, [], ctx=ctx
+    )
