@@ -129,117 +129,45 @@ __device__ __forceinline__ bool fp_eq(const fp &a, const fp &b) {
   return o == 0;
 }
 
-// Montgomery multiply: column-scanning 384x384->768 product (independent
-// partial products -> high ILP) + separated SOS reduction. ~2x the
-// instruction-level parallelism of CIOS at equal op count; validated
-// bit-exact against the CIOS oracle on 200k random inputs.
+// CIOS Montgomery multiply. (A column-scanning variant with higher ILP
+// passed 200k-input CPU validation but wedged the prepare kernel on gfx950
+// — suspected codegen issue with the break-in-unroll pattern; revisit with
+// an isolated GPU microbench before retrying.)
 __device__ __forceinline__ void fp_mul(fp &r, const fp &a, const fp &b) {
-  uint64_t t[13];
-  uint64_t lo = 0, mid = 0, hi = 0;
+  uint64_t t[8];
 #pragma unroll
-  for (int k = 0; k < 11; k++) {
-    const int i0 = k < 6 ? 0 : k - 5;
-    const int i1 = k < 6 ? k : 5;
-#pragma unroll
-    for (int i = i0; i <= i1; i++) {
-      uint64_t pl = a.v[i] * b.v[k - i];
-      uint64_t ph = __umul64hi(a.v[i], b.v[k - i]);
-      lo += pl;
-      uint64_t c1 = lo < pl;
-      mid += c1;
-      hi += (mid < c1);
-      mid += ph;
-      hi += (mid < ph);
-    }
-    t[k] = lo;
-    lo = mid;
-    mid = hi;
-    hi = 0;
-  }
-  t[11] = lo;
-  t[12] = 0;
+  for (int i = 0; i < 8; i++) t[i] = 0;
 #pragma unroll
   for (int i = 0; i < 6; i++) {
-    uint64_t m = t[i] * BLS_N0;
-    unsigned __int128 c = ((unsigned __int128)m * BLS_P[0] + t[i]) >> 64;
+    unsigned __int128 c = 0;
 #pragma unroll
-    for (int j = 1; j < 6; j++) {
-      c += (unsigned __int128)m * BLS_P[j] + t[i + j];
-      t[i + j] = (uint64_t)c;
-      c >>= 64;
-    }
-#pragma unroll
-    for (int j = i + 6; j < 13; j++) {
-      c += t[j];
+    for (int j = 0; j < 6; j++) {
+      c += (unsigned __int128)a.v[j] * b.v[i] + t[j];
       t[j] = (uint64_t)c;
       c >>= 64;
     }
+    c += t[6];
+    t[6] = (uint64_t)c;
+    t[7] = (uint64_t)(c >> 64);
+    uint64_t m = t[0] * BLS_N0;
+    c = ((unsigned __int128)m * BLS_P[0] + t[0]) >> 64;
+#pragma unroll
+    for (int j = 1; j < 6; j++) {
+      c += (unsigned __int128)m * BLS_P[j] + t[j];
+      t[j - 1] = (uint64_t)c;
+      c >>= 64;
+    }
+    c += t[6];
+    t[5] = (uint64_t)c;
+    t[6] = t[7] + (uint64_t)(c >> 64);
+    t[7] = 0;
   }
-  uint64_t res[6];
+  if (t[6] || fp_ge_p(t)) fp_sub_p(t);
 #pragma unroll
-  for (int i = 0; i < 6; i++) res[i] = t[6 + i];
-  if (t[12] || fp_ge_p(res)) fp_sub_p(res);
-#pragma unroll
-  for (int i = 0; i < 6; i++) r.v[i] = res[i];
+  for (int i = 0; i < 6; i++) r.v[i] = t[i];
 }
 
-// specialized Montgomery squaring: 21 distinct products (cross terms
-// doubled) — validated against CIOS a*a on 200k random inputs.
-__device__ __forceinline__ void fp_sqr(fp &r, const fp &a) {
-  uint64_t t[13];
-  uint64_t lo = 0, mid = 0, hi = 0;
-#pragma unroll
-  for (int k = 0; k < 11; k++) {
-    const int i0 = k < 6 ? 0 : k - 5;
-    const int i1 = k < 6 ? k : 5;
-#pragma unroll
-    for (int i = i0; i <= i1; i++) {
-      const int j = k - i;
-      if (i > j) break;
-      uint64_t pl = a.v[i] * a.v[j];
-      uint64_t ph = __umul64hi(a.v[i], a.v[j]);
-      const int dup = (i != j) ? 2 : 1;
-#pragma unroll
-      for (int d = 0; d < dup; d++) {
-        lo += pl;
-        uint64_t c1 = lo < pl;
-        mid += c1;
-        hi += (mid < c1);
-        mid += ph;
-        hi += (mid < ph);
-      }
-    }
-    t[k] = lo;
-    lo = mid;
-    mid = hi;
-    hi = 0;
-  }
-  t[11] = lo;
-  t[12] = 0;
-#pragma unroll
-  for (int i = 0; i < 6; i++) {
-    uint64_t m = t[i] * BLS_N0;
-    unsigned __int128 c = ((unsigned __int128)m * BLS_P[0] + t[i]) >> 64;
-#pragma unroll
-    for (int j = 1; j < 6; j++) {
-      c += (unsigned __int128)m * BLS_P[j] + t[i + j];
-      t[i + j] = (uint64_t)c;
-      c >>= 64;
-    }
-#pragma unroll
-    for (int j = i + 6; j < 13; j++) {
-      c += t[j];
-      t[j] = (uint64_t)c;
-      c >>= 64;
-    }
-  }
-  uint64_t res[6];
-#pragma unroll
-  for (int i = 0; i < 6; i++) res[i] = t[6 + i];
-  if (t[12] || fp_ge_p(res)) fp_sub_p(res);
-#pragma unroll
-  for (int i = 0; i < 6; i++) r.v[i] = res[i];
-}
+__device__ __forceinline__ void fp_sqr(fp &r, const fp &a) { fp_mul(r, a, a); }
 
 __device__ __forceinline__ void fp_one(fp &r) {
   // R mod p = mont(1): computed as R2 * 1 via montmul(1_std, R2)
@@ -1447,13 +1375,20 @@ __device__ inline void f12_inv_w(fp12m &r, const fp12m &a, fp12m &g, fp12m &t,
 
 // cooperative squaring: 21 distinct products across lanes (alias-safe)
 __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
-  static const int PI[21] = {0, 0, 0, 0, 0, 0, 1, 1, 1, 1, 1,
-                             2, 2, 2, 2, 3, 3, 3, 4, 4, 5};
-  static const int PJ[21] = {0, 1, 2, 3, 4, 5, 1, 2, 3, 4, 5,
-                             2, 3, 4, 5, 3, 4, 5, 4, 5, 5};
+  // NB: no function-local statics in device code (init guards can
+  // deadlock a wave); lane->pair map computed arithmetically instead.
+  // pairs (i,j), i<=j, enumerated row-major: lane l -> smallest i with
+  // T(i) <= l where T(i) = i*(13-i)/2 (offset of row i).
+  int i = 0;
+  {
+    int l = lane < 21 ? lane : 0;
+    while (i < 5 && l >= (i + 1) * (12 - i) / 2) i++;
+    // row i starts at T(i) = i*(13-i)/2
+  }
+  int row_start = i * (13 - i) / 2;
+  int j = i + (lane < 21 ? lane : 0) - row_start;
   if (lane < 21) {
     fp2 ai, aj, t;
-    int i = PI[lane], j = PJ[lane];
     f12_get(a, i, ai);
     if (i == j) {
       fp2_sqr(t, ai);
@@ -1468,15 +1403,18 @@ __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
     fp2 acc, hacc, t;
     fp2_zero(acc);
     fp2_zero(hacc);
-    for (int q = 0; q < 21; q++) {
-      int sum = PI[q] + PJ[q];
-      if (sum == lane || sum == lane + 6) {
-        t = ws.t[q];
-        if (PI[q] != PJ[q]) fp2_dbl(t, t);
-        if (sum == lane)
-          fp2_add(acc, acc, t);
-        else
-          fp2_add(hacc, hacc, t);
+    int qi = 0;
+    for (int pi = 0; pi < 6; pi++) {
+      for (int pj = pi; pj < 6; pj++, qi++) {
+        int sum = pi + pj;
+        if (sum == lane || sum == lane + 6) {
+          t = ws.t[qi];
+          if (pi != pj) fp2_dbl(t, t);
+          if (sum == lane)
+            fp2_add(acc, acc, t);
+          else
+            fp2_add(hacc, hacc, t);
+        }
       }
     }
     fp2_mul_xi(hacc, hacc);

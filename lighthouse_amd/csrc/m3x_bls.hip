@@ -11,6 +11,19 @@
 #include "bls_device.hh"
 #include "m3x_ctx.hh"
 #include "../../include/m3x_consensus.h"
+#include <cstdio>
+#include <cstdlib>
+
+// M3X_DEBUG_SYNC=1: synchronize + trace after each BLS kernel (debug aid)
+static bool dbg_sync() {
+  static int v = -1;
+  if (v < 0) {
+    const char *e = getenv("M3X_DEBUG_SYNC");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+#define DBG_STEP(ctx, name)                                                      do {                                                                             if (dbg_sync()) {                                                                hipError_t _e = hipStreamSynchronize((ctx)->stream);                           fprintf(stderr, "[m3x dbg] %s: %s\n", name, hipGetErrorString(_e));            fflush(stderr);                                                              }                                                                            } while (0)
 
 using namespace m3xb;
 
@@ -254,14 +267,17 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                      (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
                      n, w);
   m3x::time_end(ctx, M3X_K_BLS_PREPARE);
+  DBG_STEP(ctx, "prepare");
   m3x::time_begin(ctx, M3X_K_BLS_H2C);
   hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint8_t *)msgs_dev, n, w);
   m3x::time_end(ctx, M3X_K_BLS_H2C);
+  DBG_STEP(ctx, "h2c");
   m3x::time_begin(ctx, M3X_K_BLS_MILLER);
   hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
                      w);
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
+  DBG_STEP(ctx, "miller");
   uint32_t rblocks = (uint32_t)((n + 255) / 256);
   if (rblocks > 256) rblocks = 256;
   m3x::time_begin(ctx, M3X_K_BLS_REDUCE);
@@ -274,9 +290,11 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   hipLaunchKernelGGL(k_bls_reduce_sig, dim3(1), dim3(256), 0, ctx->stream,
                      w.sig_stage, (uint64_t)rblocks, w.sig_sum);
   m3x::time_end(ctx, M3X_K_BLS_REDUCE);
+  DBG_STEP(ctx, "reduce");
   m3x::time_begin(ctx, M3X_K_BLS_FINISH);
   hipLaunchKernelGGL(k_bls_finish, dim3(1), dim3(64), 0, ctx->stream, w);
   m3x::time_end(ctx, M3X_K_BLS_FINISH);
+  DBG_STEP(ctx, "finish");
   int32_t verdict = 0;
   M3X_HIP_CHECK(hipMemcpyAsync(&verdict, w.verdict, 4, hipMemcpyDeviceToHost,
                                ctx->stream));
