@@ -104,6 +104,17 @@ int32_t m3x_merkleize_batch(m3x_ctx *ctx, const uint8_t *chunks,
                             const uint32_t *offsets, uint64_t n_elems,
                             uint8_t *out_roots);
 
+/* swap-or-not committee shuffle — restated from
+ * consensus/swap_or_not_shuffle/src/shuffle_list.rs; in-place on u32
+ * indices; rounds = ChainSpec::shuffle_round_count (90 mainnet,
+ * chain_spec.rs:632); forwards semantics as the reference. */
+int32_t m3x_shuffle_list(m3x_ctx *ctx, uint32_t *indices, uint64_t list_size,
+                         uint32_t rounds, const uint8_t seed[32],
+                         int32_t forwards);
+int32_t m3x_shuffle_list_dev(m3x_ctx *ctx, void *indices_dev,
+                             uint64_t list_size, uint32_t rounds,
+                             const uint8_t seed[32], int32_t forwards);
+
 /* ---- BLS12-381 batched signature-set verification (hot path #1) ---- */
 
 /* key_validate a batch of compressed pubkeys: decompress + infinity reject +

@@ -84,6 +84,28 @@ def _bind(lib):
         ),
         ("m3x_timing_enable", [ctypes.c_void_p, ctypes.c_int32]),
         (
+            "m3x_shuffle_list",
+            [
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_uint64,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+                ctypes.c_int32,
+            ],
+        ),
+        (
+            "m3x_shuffle_list_dev",
+            [
+                ctypes.c_void_p,
+                ctypes.c_void_p,
+                ctypes.c_uint64,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+                ctypes.c_int32,
+            ],
+        ),
+        (
             "m3x_merkleize_batch",
             [
                 ctypes.c_void_p,

@@ -74,6 +74,12 @@ void m3x_oracle_root_vector_root(const uint8_t *roots, uint64_t n,
 void m3x_oracle_root_list_root(const uint8_t *roots, uint64_t n,
                                uint64_t limit, uint8_t out[32]);
 
+/* swap-or-not shuffle (shuffle_list.rs restatement; §8f.1). In-place on
+ * u32 indices; 0 ok, -1 on the reference's None conditions. */
+int m3x_oracle_shuffle_list(uint32_t *input, uint64_t list_size,
+                            uint8_t rounds, const uint8_t seed[32],
+                            int forwards);
+
 #ifdef __cplusplus
 }
 #endif
