@@ -1420,19 +1420,15 @@ __device__ inline void f12_inv_w(fp12m &r, const fp12m &a, fp12m &g, fp12m &t,
 }
 
 // cooperative squaring: 21 distinct products across lanes (alias-safe)
+__constant__ int F12_SQR_PI[21] = {0, 0, 0, 0, 0, 0, 1, 1, 1, 1, 1,
+                                   2, 2, 2, 2, 3, 3, 3, 4, 4, 5};
+__constant__ int F12_SQR_PJ[21] = {0, 1, 2, 3, 4, 5, 1, 2, 3, 4, 5,
+                                   2, 3, 4, 5, 3, 4, 5, 4, 5, 5};
+
 __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
-  // NB: no function-local statics in device code (init guards can
-  // deadlock a wave); lane->pair map computed arithmetically instead.
-  // pairs (i,j), i<=j, enumerated row-major: lane l -> smallest i with
-  // T(i) <= l where T(i) = i*(13-i)/2 (offset of row i).
-  int i = 0;
-  {
-    int l = lane < 21 ? lane : 0;
-    while (i < 5 && l >= (i + 1) * (12 - i) / 2) i++;
-    // row i starts at T(i) = i*(13-i)/2
-  }
-  int row_start = i * (13 - i) / 2;
-  int j = i + (lane < 21 ? lane : 0) - row_start;
+  int li = lane < 21 ? lane : 0;
+  int i = F12_SQR_PI[li];
+  int j = F12_SQR_PJ[li];
   if (lane < 21) {
     fp2 ai, aj, t;
     f12_get(a, i, ai);

@@ -20,6 +20,8 @@ enum m3x_kernel_id {
 struct m3x_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
+  hipStream_t stream2 = nullptr; // overlap of independent kernels
+  hipEvent_t ev_s2 = nullptr;
   // device-resident zero-hash ladder Z[0..64] (computed on GPU at create)
   uint8_t *zeros_dev = nullptr;
   // growable scratch buffers for merkle reduction stages

@@ -290,7 +290,10 @@ int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
     delete ctx;
     return M3X_ERR_HIP;
   }
-  if (hipStreamCreate(&ctx->stream) != hipSuccess) {
+  if (hipStreamCreate(&ctx->stream) != hipSuccess ||
+      hipStreamCreate(&ctx->stream2) != hipSuccess ||
+      hipEventCreateWithFlags(&ctx->ev_s2, hipEventDisableTiming) !=
+          hipSuccess) {
     delete ctx;
     return M3X_ERR_HIP;
   }
@@ -320,6 +323,8 @@ void m3x_ctx_destroy(m3x_ctx *ctx) {
   if (ctx->scratch_a) (void)hipFree(ctx->scratch_a);
   if (ctx->scratch_b) (void)hipFree(ctx->scratch_b);
   if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
+  if (ctx->stream2) (void)hipStreamDestroy(ctx->stream2);
+  if (ctx->ev_s2) (void)hipEventDestroy(ctx->ev_s2);
   delete ctx;
 }
 
