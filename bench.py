@@ -243,7 +243,6 @@ def main():
     if world > 1:
         dist.barrier()
     torch.cuda.synchronize()
-    ctx.timing_enable(True)
     t0 = time.time()
     for _ in range(args.steps):
         v = bls_step()
@@ -257,6 +256,12 @@ def main():
         t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
+    # separate instrumented pass (per-kernel timing serializes the
+    # prepare/h2c stream overlap, so it runs OUTSIDE the timed region)
+    ctx.timing_enable(True)
+    bls_step()
+    merkle_step()
+    ctx.timing_enable(False)
     ktimes = ctx.kernel_times()
 
     # C4: synthetic Deneb block import (131 sets: 128 aggregates k=512 +
@@ -406,7 +411,7 @@ def main():
                 "state_root_node_hashes": full_nodes,
                 "bls_batch_ms": bls_only * 1e3,
                 "kernel_ms_per_step": {
-                    k: v[0] / max(steps, 1) for k, v in ktimes.items() if v[1]
+                    k: v[0] for k, v in ktimes.items() if v[1]
                 },
                 "c4_block_import_ms": c4_ms,
                 "cpu_sha_baseline": cpu_sha,

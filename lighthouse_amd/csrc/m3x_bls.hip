@@ -270,12 +270,12 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   DBG_STEP(ctx, "prepare");
   // h2c is independent of prepare: run it on the second stream so the two
   // ~1-wave/SIMD kernels co-reside (both fit at 2 waves/SIMD by VGPR count)
-  m3x::time_begin(ctx, M3X_K_BLS_H2C);
+  m3x::time_begin_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
   hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream2,
                      (const uint8_t *)msgs_dev, n, w);
+  m3x::time_end_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
   M3X_HIP_CHECK(hipEventRecord(ctx->ev_s2, ctx->stream2));
   M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream, ctx->ev_s2, 0));
-  m3x::time_end(ctx, M3X_K_BLS_H2C);
   DBG_STEP(ctx, "h2c");
   m3x::time_begin(ctx, M3X_K_BLS_MILLER);
   hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,

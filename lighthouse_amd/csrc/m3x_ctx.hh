@@ -47,6 +47,8 @@ namespace m3x {
 // wrap a launch with events when ctx->timing (call after the launch)
 void time_begin(m3x_ctx *ctx, int k);
 void time_end(m3x_ctx *ctx, int k);
+void time_begin_s(m3x_ctx *ctx, int k, hipStream_t st);
+void time_end_s(m3x_ctx *ctx, int k, hipStream_t st);
 } // namespace m3x
 
 #define M3X_HIP_CHECK(expr)                                                    \

@@ -208,24 +208,27 @@ __global__ void k_finalize(const uint8_t *__restrict__ node_in,
 // ---------------------------------------------------------------- host side
 
 namespace m3x {
-void time_begin(m3x_ctx *ctx, int k) {
+void time_begin_s(m3x_ctx *ctx, int k, hipStream_t st) {
   if (!ctx->timing) return;
   if (!ctx->ev_a[k]) {
     (void)hipEventCreate(&ctx->ev_a[k]);
     (void)hipEventCreate(&ctx->ev_b[k]);
   }
-  (void)hipEventRecord(ctx->ev_a[k], ctx->stream);
+  (void)hipEventRecord(ctx->ev_a[k], st);
 }
 
-void time_end(m3x_ctx *ctx, int k) {
+void time_end_s(m3x_ctx *ctx, int k, hipStream_t st) {
   if (!ctx->timing) return;
-  (void)hipEventRecord(ctx->ev_b[k], ctx->stream);
+  (void)hipEventRecord(ctx->ev_b[k], st);
   (void)hipEventSynchronize(ctx->ev_b[k]);
   float ms = 0;
   (void)hipEventElapsedTime(&ms, ctx->ev_a[k], ctx->ev_b[k]);
   ctx->kernel_ms[k] += ms;
   ctx->kernel_launches[k]++;
 }
+
+void time_begin(m3x_ctx *ctx, int k) { time_begin_s(ctx, k, ctx->stream); }
+void time_end(m3x_ctx *ctx, int k) { time_end_s(ctx, k, ctx->stream); }
 
 int ensure_scratch(m3x_ctx *ctx, uint8_t **buf, uint64_t *cur,
                    uint64_t bytes) {
