@@ -1070,75 +1070,8 @@ __device__ inline bool f12_is_one(const fp12m &a) {
   return true;
 }
 
-// f^p: c_k -> conj(c_k)*FW1^k ; f^(p^2): c_k -> c_k*FW2^k (alias-safe)
-__device__ inline void f12_frob(fp12m &r, const fp12m &a, int power) {
-  fp2 fw1, fw;
-  FP2_LOAD_C(fw1, FROB_W1);
-  if (power == 2) {
-    fp2 c1;
-    fp2_conj(c1, fw1);
-    fp2_mul(fw, fw1, c1); // FW2 = norm(FW1), lands in Fp
-  } else {
-    fw = fw1;
-  }
-  fp2 accw;
-  fp2_one(accw);
-  for (int k = 0; k < 6; k++) {
-    fp2 c;
-    f12_get(a, k, c);
-    if (power == 1) fp2_conj(c, c);
-    fp2_mul(c, c, accw);
-    f12_set(r, k, c);
-    fp2_mul(accw, accw, fw);
-  }
-}
 
-// r = a^-1 via the sigma-conjugate product; r, g, t distinct from a and
-// from each other.
-__device__ inline void f12_inv_nn(fp12m &r, const fp12m &a, fp12m &g,
-                                  fp12m &t) {
-  fp2 z6;
-  FP2_LOAD_C(z6, ZETA6);
-  f12_one(g);
-  for (int i = 1; i < 6; i++) {
-    fp2 zi, accw;
-    zi = z6;
-    for (int q = 1; q < i; q++) fp2_mul(zi, zi, z6);
-    fp2_one(accw);
-    for (int j = 0; j < 6; j++) {
-      fp2 c;
-      f12_get(a, j, c);
-      fp2_mul(c, c, accw);
-      f12_set(t, j, c);
-      fp2_mul(accw, accw, zi);
-    }
-    f12_mul_nn(r, g, t);
-    f12_copy(g, r);
-  }
-  f12_mul_nn(t, a, g); // norm (lands in Fp2)
-  fp2 n0, ninv;
-  f12_get(t, 0, n0);
-  fp2_inv(ninv, n0);
-  for (int j = 0; j < 6; j++) {
-    fp2 c;
-    f12_get(g, j, c);
-    fp2_mul(c, c, ninv);
-    f12_set(r, j, c);
-  }
-}
 
-// r = a^|x| (|x| = BLS_X_ABS, hamming 6); r, t distinct from a and each
-// other. In the cyclotomic subgroup a^x (x<0) = conj6(a^|x|).
-__device__ inline void f12_pow_xabs_nn(fp12m &r, const fp12m &a, fp12m &t) {
-  f12_copy(r, a);
-  for (int b = 62; b >= 0; b--) {
-    f12_sqr_nn(t, r);
-    if ((BLS_X_ABS >> b) & 1)
-      f12_mul_nn(r, t, a);
-    else
-      f12_copy(r, t);
-  }
-}
 
 // ---------------------------------------------------------------- pairing ---
 
@@ -1230,45 +1163,6 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
   if (cur != &out) f12_copy(out, *cur);
 }
 
-// final exponentiation: easy part + HHT chain
-// hard = (x-1)^2 (x+p) (x^2+p^2-1) + 3  (== 3*(p^4-p^2+1)/r; the cube is
-// harmless for the ==1 test since gcd(3,r)=1 — validated in Python).
-// r and f must be distinct from each other and from the 6 scratch slots.
-__device__ inline void final_exp_s(fp12m &r, const fp12m &f, fp12m *s) {
-  f12_copy(s[0], f);
-  f12_conj6_ip(s[0]); // f^(p^6)
-  f12_inv_nn(s[1], f, s[2], s[3]);
-  f12_mul_nn(s[2], s[0], s[1]); // f^(p^6-1)
-  f12_frob(s[0], s[2], 2);
-  f12_mul_nn(s[1], s[0], s[2]); // e = f^((p^6-1)(p^2+1)), cyclotomic
-  // u = e^(x-1) = conj6(e^|x| * e)
-  f12_pow_xabs_nn(s[0], s[1], s[3]);
-  f12_mul_nn(s[2], s[0], s[1]);
-  f12_conj6_ip(s[2]); // e^(x-1)
-  // v = u^(x-1)
-  f12_pow_xabs_nn(s[0], s[2], s[3]);
-  f12_mul_nn(s[4], s[0], s[2]);
-  f12_conj6_ip(s[4]); // e^((x-1)^2)
-  // w1 = conj6(v^|x|) * v^p
-  f12_pow_xabs_nn(s[0], s[4], s[3]);
-  f12_conj6_ip(s[0]);
-  f12_frob(s[2], s[4], 1);
-  f12_mul_nn(s[5], s[0], s[2]); // w1
-  // w2 = (w1^x)^x * w1^(p^2) * conj6(w1)
-  f12_pow_xabs_nn(s[0], s[5], s[3]);
-  f12_conj6_ip(s[0]);
-  f12_pow_xabs_nn(s[2], s[0], s[3]);
-  f12_conj6_ip(s[2]); // w1^(x^2)
-  f12_frob(s[0], s[5], 2);
-  f12_mul_nn(s[4], s[2], s[0]);
-  f12_copy(s[0], s[5]);
-  f12_conj6_ip(s[0]);
-  f12_mul_nn(s[2], s[4], s[0]); // w2
-  // r = w2 * e^3 (e still in s[1])
-  f12_sqr_nn(s[0], s[1]);
-  f12_mul_nn(s[3], s[0], s[1]);
-  f12_mul_nn(r, s[2], s[3]);
-}
 
 // ------------------------- wave-cooperative fp12 (64-thread finish path) ---
 // All 64 lanes execute the same control flow; fp12 multiplies fan the 36

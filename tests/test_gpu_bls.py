@@ -208,3 +208,39 @@ def test_single_and_fast_aggregate_verify(ctx, oracle):
     assert bls.eth_fast_aggregate_verify(
         bls.Signature.infinity(), b"\x00" * 32, [], ctx=ctx
     )
+
+
+def test_aggregate_pubkey_at_infinity_rejected(ctx, oracle):
+    """a set whose pubkeys sum to the identity (pk + (-pk)) must verify
+    False (blst BLST_PK_IS_INFINITY semantics; DESIGN.md boundary rules) —
+    on both the GPU path and the oracle."""
+    import ctypes as ct
+
+    from lighthouse_amd import bls
+
+    pk_unc = bytes.fromhex(FIXTURES["interop"][0]["pk_uncompressed_hex"])
+    # -pk = [r-1] pk via the oracle's scalar-mult helper
+    r_minus_1 = (
+        0x73EDA753299D7D483339D80809A1D80553BDA402FFFE5BFEFFFFFFFF00000001 - 1
+    ).to_bytes(32, "big")
+    neg = ct.create_string_buffer(96)
+    assert oracle.m3x_oracle_bls_g1_mul(pk_unc, r_minus_1, neg) == 0
+    sig = bls.Signature.from_compressed(
+        bytes.fromhex(FIXTURES["signatures"][0]["sig_compressed_hex"])
+    )
+    msg = bytes.fromhex(FIXTURES["signatures"][0]["msg_hex"])
+    s = bls.SignatureSet(
+        sig,
+        [bls.PublicKey.from_uncompressed(pk_unc),
+         bls.PublicKey.from_uncompressed(neg.raw)],
+        msg,
+    )
+    assert bls.verify_signature_sets([s], ctx=ctx) is False
+    offs = (ct.c_uint32 * 2)(0, 2)
+    rnd = (ct.c_uint64 * 1)(12345)
+    assert (
+        oracle.m3x_oracle_bls_verify_sets(
+            msg, sig.serialize(), pk_unc + neg.raw, offs, rnd, 1
+        )
+        == 0
+    )
