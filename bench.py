@@ -306,6 +306,30 @@ def main():
         c4_ms = (time.time() - tb) * 1e3
         assert v == 1
 
+    # incremental registry cache: per-block delta rehash
+    incr_ms = None
+    if rank == 0 and world == 1:
+        from lighthouse_amd import tree_hash as th
+
+        cache = th.RegistryCache(ssz_full, N_VALIDATORS, ctx=ctx)
+        import random as _rnd
+
+        rng = _rnd.Random(5)
+        idxs = sorted(rng.sample(range(N_VALIDATORS), 2048))
+        blob = b"".join(
+            ssz_full[121 * i : 121 * i + 80]
+            + int(31_000_000_000).to_bytes(8, "little")
+            + ssz_full[121 * i + 88 : 121 * (i + 1)]
+            for i in idxs
+        )
+        cache.update(idxs, blob)  # warm
+        torch.cuda.synchronize()
+        tb = time.time()
+        cache.update(idxs, blob)
+        torch.cuda.synchronize()
+        incr_ms = (time.time() - tb) * 1e3
+        cache.close()
+
     # split timing: one more pass of each, timed separately (for extras)
     torch.cuda.synchronize()
     tb = time.time()
@@ -414,6 +438,7 @@ def main():
                     k: v[0] for k, v in ktimes.items() if v[1]
                 },
                 "c4_block_import_ms": c4_ms,
+                "registry_incremental_update_2048_ms": incr_ms,
                 "cpu_sha_baseline": cpu_sha,
             },
             "roofline": roofline,

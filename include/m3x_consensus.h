@@ -96,6 +96,27 @@ int32_t m3x_validator_subtree_root_dev(m3x_ctx *ctx, const void *ssz_dev,
                                        uint64_t n, uint32_t depth,
                                        uint8_t out_root[32]);
 
+/* ---- incremental registry merkleize (SURVEY §8f.3 — the milhouse-style
+ * cached rehash: beacon_state.rs:1990-2021 has_pending_updates path).
+ * The cache holds every tree level in HBM; an update rehashes only the
+ * dirty leaves and their root paths. ---- */
+typedef struct m3x_registry_cache m3x_registry_cache;
+
+/* build the cache from n packed 121-byte records; capacity is the next
+ * power of two >= max(n, 256) (appends beyond it are an error this round). */
+int32_t m3x_registry_cache_create(m3x_ctx *ctx, const uint8_t *ssz,
+                                  uint64_t n, m3x_registry_cache **out);
+void m3x_registry_cache_destroy(m3x_registry_cache *cache);
+/* current List[Validator, 2^40] root (zero-cap + mix_in_length) */
+int32_t m3x_registry_cache_root(m3x_ctx *ctx, m3x_registry_cache *cache,
+                                uint8_t out_root[32]);
+/* apply m updated/appended records at `indices` (each < capacity;
+ * n grows to max over indices+1) and return the new root */
+int32_t m3x_registry_cache_update(m3x_ctx *ctx, m3x_registry_cache *cache,
+                                  const uint64_t *indices,
+                                  const uint8_t *recs /* m*121 */, uint64_t m,
+                                  uint8_t out_root[32]);
+
 /* Batched small-container merkleize: element i owns chunks
  * [offsets[i], offsets[i+1]) (<=32 chunks each); out_roots[i] = its
  * hash_tree_root (depth = ceil_log2(count)). One launch for the

@@ -84,6 +84,22 @@ def _bind(lib):
         ),
         ("m3x_timing_enable", [ctypes.c_void_p, ctypes.c_int32]),
         (
+            "m3x_registry_cache_create",
+            [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint64,
+             ctypes.POINTER(ctypes.c_void_p)],
+        ),
+        ("m3x_registry_cache_destroy", [ctypes.c_void_p]),
+        (
+            "m3x_registry_cache_root",
+            [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_char_p],
+        ),
+        (
+            "m3x_registry_cache_update",
+            [ctypes.c_void_p, ctypes.c_void_p,
+             ctypes.POINTER(ctypes.c_uint64), ctypes.c_char_p,
+             ctypes.c_uint64, ctypes.c_char_p],
+        ),
+        (
             "m3x_shuffle_list",
             [
                 ctypes.c_void_p,
@@ -175,7 +191,9 @@ def _bind(lib):
         except AttributeError:
             continue
         fn.argtypes = argtypes
-        fn.restype = ctypes.c_int32
+        fn.restype = (
+            None if name == "m3x_registry_cache_destroy" else ctypes.c_int32
+        )
     return lib
 
 

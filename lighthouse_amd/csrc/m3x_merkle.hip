@@ -631,3 +631,267 @@ extern "C" int32_t m3x_merkleize_batch(m3x_ctx *ctx, const uint8_t *chunks,
   if (!pooled) (void)hipFree(base);
   return rc;
 }
+
+// ------------------- incremental registry merkleize --------
+// The cache stores all tree levels contiguously in HBM (level l at
+// offs[l], counts cap>>l). Updates rehash dirty leaves + root paths only
+// (identical concurrent writes to a shared parent are benign).
+
+struct m3x_registry_cache {
+  uint32_t cap_log2 = 0;
+  uint64_t capacity = 0;
+  uint64_t n = 0; // current list length (mix_in_length)
+  uint8_t *levels = nullptr;   // sum_{l=0..cap_log2} (cap>>l) * 32 bytes
+  uint64_t *dirty_a = nullptr; // ping-pong dirty index lists
+  uint64_t *dirty_b = nullptr;
+  uint8_t *recs = nullptr;     // staging for update records
+  uint64_t recs_cap = 0;
+};
+
+namespace {
+
+__device__ __forceinline__ uint64_t lvl_off(uint32_t cap_log2, uint32_t l) {
+  // sum of counts of levels < l, in nodes: cap*(2 - 2^-(l-1)) ... computed
+  // iteratively (l <= 24)
+  uint64_t off = 0, c = 1ull << cap_log2;
+  for (uint32_t q = 0; q < l; q++) {
+    off += c;
+    c >>= 1;
+  }
+  return off;
+}
+
+// hash one level range into the next (full build)
+__global__ void k_cache_level(const uint8_t *__restrict__ src, uint64_t n_out,
+                              uint8_t *__restrict__ dst) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n_out) return;
+  uint32_t l[8], r[8], o[8];
+  const uint32_t *a = reinterpret_cast<const uint32_t *>(src + 64 * i);
+#pragma unroll
+  for (int q = 0; q < 8; q++) l[q] = __builtin_bswap32(a[q]);
+#pragma unroll
+  for (int q = 0; q < 8; q++) r[q] = __builtin_bswap32(a[8 + q]);
+  m3x::sha256_node(l, r, o);
+  uint32_t *d = reinterpret_cast<uint32_t *>(dst + 32 * i);
+#pragma unroll
+  for (int q = 0; q < 8; q++) d[q] = __builtin_bswap32(o[q]);
+}
+
+// recompute leaf roots for m dirty records and scatter into level 0
+__global__ void k_cache_update_leaves(const uint8_t *__restrict__ recs,
+                                      const uint64_t *__restrict__ idx,
+                                      uint64_t m, uint8_t *__restrict__ level0) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= m) return;
+  const uint8_t *v = recs + 121 * t;
+  uint32_t c[8][8];
+#pragma unroll
+  for (int i = 0; i < 8; i++)
+#pragma unroll
+    for (int j = 0; j < 8; j++) c[i][j] = 0;
+  {
+    uint32_t l[8], r[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++) l[i] = be_load_u8x4(v + 4 * i);
+#pragma unroll
+    for (int i = 0; i < 4; i++) r[i] = be_load_u8x4(v + 32 + 4 * i);
+#pragma unroll
+    for (int i = 4; i < 8; i++) r[i] = 0;
+    m3x::sha256_node(l, r, c[0]);
+  }
+#pragma unroll
+  for (int i = 0; i < 8; i++) c[1][i] = be_load_u8x4(v + 48 + 4 * i);
+  c[2][0] = be_load_u8x4(v + 80);
+  c[2][1] = be_load_u8x4(v + 84);
+  c[3][0] = (uint32_t)v[88] << 24;
+  c[4][0] = be_load_u8x4(v + 89);
+  c[4][1] = be_load_u8x4(v + 93);
+  c[5][0] = be_load_u8x4(v + 97);
+  c[5][1] = be_load_u8x4(v + 101);
+  c[6][0] = be_load_u8x4(v + 105);
+  c[6][1] = be_load_u8x4(v + 109);
+  c[7][0] = be_load_u8x4(v + 113);
+  c[7][1] = be_load_u8x4(v + 117);
+  uint32_t h01[8], h23[8], h45[8], h67[8], ha[8], hb[8], root[8];
+  m3x::sha256_node(c[0], c[1], h01);
+  m3x::sha256_node(c[2], c[3], h23);
+  m3x::sha256_node(c[4], c[5], h45);
+  m3x::sha256_node(c[6], c[7], h67);
+  m3x::sha256_node(h01, h23, ha);
+  m3x::sha256_node(h45, h67, hb);
+  m3x::sha256_node(ha, hb, root);
+  uint32_t *d = reinterpret_cast<uint32_t *>(level0 + 32 * idx[t]);
+#pragma unroll
+  for (int q = 0; q < 8; q++) d[q] = __builtin_bswap32(root[q]);
+}
+
+// one propagation step: for each dirty node at level l, recompute its
+// parent at level l+1 and emit the parent index
+__global__ void k_cache_propagate(uint8_t *__restrict__ levels,
+                                  uint32_t cap_log2, uint32_t l,
+                                  const uint64_t *__restrict__ dirty,
+                                  uint64_t m,
+                                  uint64_t *__restrict__ dirty_next) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= m) return;
+  uint64_t child = dirty[t];
+  uint64_t parent = child >> 1;
+  const uint8_t *src =
+      levels + 32 * (lvl_off(cap_log2, l) + (parent << 1));
+  uint8_t *dst = levels + 32 * (lvl_off(cap_log2, l + 1) + parent);
+  uint32_t a[8], b[8], o[8];
+  const uint32_t *s32 = reinterpret_cast<const uint32_t *>(src);
+#pragma unroll
+  for (int q = 0; q < 8; q++) a[q] = __builtin_bswap32(s32[q]);
+#pragma unroll
+  for (int q = 0; q < 8; q++) b[q] = __builtin_bswap32(s32[8 + q]);
+  m3x::sha256_node(a, b, o);
+  uint32_t *d32 = reinterpret_cast<uint32_t *>(dst);
+#pragma unroll
+  for (int q = 0; q < 8; q++) d32[q] = __builtin_bswap32(o[q]);
+  dirty_next[t] = parent;
+}
+
+} // namespace
+
+extern "C" {
+
+void m3x_registry_cache_destroy(m3x_registry_cache *c) {
+  if (!c) return;
+  if (c->levels) (void)hipFree(c->levels);
+  if (c->dirty_a) (void)hipFree(c->dirty_a);
+  if (c->dirty_b) (void)hipFree(c->dirty_b);
+  if (c->recs) (void)hipFree(c->recs);
+  delete c;
+}
+
+int32_t m3x_registry_cache_create(m3x_ctx *ctx, const uint8_t *ssz,
+                                  uint64_t n, m3x_registry_cache **out) {
+  if (!ctx || !out || n > (1ull << 24)) return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  auto *c = new m3x_registry_cache();
+  uint64_t cap = 256;
+  uint32_t cl = 8;
+  while (cap < n) {
+    cap <<= 1;
+    cl++;
+  }
+  c->cap_log2 = cl;
+  c->capacity = cap;
+  c->n = n;
+  uint64_t total_nodes = 0, q = cap;
+  for (uint32_t l = 0; l <= cl; l++) {
+    total_nodes += q;
+    q >>= 1;
+  }
+  if (hipMalloc(&c->levels, total_nodes * 32) != hipSuccess) {
+    delete c;
+    return M3X_ERR_NOMEM;
+  }
+  // level 0: leaf roots for [0,n) then zero-ladder Z[0] (= zero chunk) pad
+  M3X_HIP_CHECK(hipMemsetAsync(c->levels, 0, cap * 32, ctx->stream));
+  if (n) {
+    void *ssz_d;
+    M3X_HIP_CHECK(hipMalloc(&ssz_d, (n * 121 + 3) & ~3ull));
+    if (hipMemcpyAsync(ssz_d, ssz, n * 121, hipMemcpyHostToDevice,
+                       ctx->stream) != hipSuccess) {
+      (void)hipFree(ssz_d);
+      m3x_registry_cache_destroy(c);
+      return M3X_ERR_HIP;
+    }
+    uint32_t blocks = (uint32_t)((n + 255) / 256);
+    hipLaunchKernelGGL(k_validator_leaves, dim3(blocks), dim3(256), 0,
+                       ctx->stream, (const uint8_t *)ssz_d, n, c->levels);
+    (void)hipStreamSynchronize(ctx->stream);
+    (void)hipFree(ssz_d);
+  }
+  // build all levels (full capacity, zero-padded leaves make the ladder)
+  uint64_t off = 0, cnt = cap;
+  for (uint32_t l = 0; l < cl; l++) {
+    uint64_t n_out = cnt >> 1;
+    uint8_t *src = c->levels + off * 32;
+    uint8_t *dst = c->levels + (off + cnt) * 32;
+    uint32_t blocks = (uint32_t)((n_out + 255) / 256);
+    hipLaunchKernelGGL(k_cache_level, dim3(blocks), dim3(256), 0, ctx->stream,
+                       src, n_out, dst);
+    off += cnt;
+    cnt = n_out;
+  }
+  // dirty lists + record staging (grown on demand)
+  if (hipMalloc(&c->dirty_a, 65536 * 8) != hipSuccess ||
+      hipMalloc(&c->dirty_b, 65536 * 8) != hipSuccess) {
+    m3x_registry_cache_destroy(c);
+    return M3X_ERR_NOMEM;
+  }
+  if (hipStreamSynchronize(ctx->stream) != hipSuccess) {
+    m3x_registry_cache_destroy(c);
+    return M3X_ERR_HIP;
+  }
+  *out = c;
+  return M3X_OK;
+}
+
+int32_t m3x_registry_cache_root(m3x_ctx *ctx, m3x_registry_cache *c,
+                                uint8_t out_root[32]) {
+  if (!ctx || !c) return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint64_t top = 0, q = c->capacity;
+  for (uint32_t l = 0; l < c->cap_log2; l++) {
+    top += q;
+    q >>= 1;
+  }
+  uint8_t *root_dev = ctx->small_pool + m3x_ctx::SMALL_POOL_IN;
+  hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream,
+                     c->levels + top * 32, c->cap_log2, 40, (int64_t)c->n,
+                     ctx->zeros_dev, root_dev);
+  M3X_HIP_CHECK(hipMemcpyAsync(out_root, root_dev, 32, hipMemcpyDeviceToHost,
+                               ctx->stream));
+  M3X_HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return M3X_OK;
+}
+
+int32_t m3x_registry_cache_update(m3x_ctx *ctx, m3x_registry_cache *c,
+                                  const uint64_t *indices,
+                                  const uint8_t *recs, uint64_t m,
+                                  uint8_t out_root[32]) {
+  if (!ctx || !c) return M3X_ERR_ARG;
+  if (m > 65536) return M3X_ERR_ARG; // dirty-list capacity this round
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  if (m) {
+    for (uint64_t i = 0; i < m; i++) {
+      if (indices[i] >= c->capacity) return M3X_ERR_ARG;
+      if (indices[i] + 1 > c->n) c->n = indices[i] + 1;
+    }
+    uint64_t need = m * 121;
+    if (c->recs_cap < need) {
+      if (c->recs) (void)hipFree(c->recs);
+      c->recs = nullptr;
+      c->recs_cap = 0;
+      if (hipMalloc(&c->recs, (need + 3) & ~3ull) != hipSuccess)
+        return M3X_ERR_NOMEM;
+      c->recs_cap = need;
+    }
+    M3X_HIP_CHECK(hipMemcpyAsync(c->recs, recs, m * 121,
+                                 hipMemcpyHostToDevice, ctx->stream));
+    M3X_HIP_CHECK(hipMemcpyAsync(c->dirty_a, indices, m * 8,
+                                 hipMemcpyHostToDevice, ctx->stream));
+    uint32_t blocks = (uint32_t)((m + 63) / 64);
+    hipLaunchKernelGGL(k_cache_update_leaves, dim3(blocks), dim3(64), 0,
+                       ctx->stream, c->recs, c->dirty_a, m, c->levels);
+    uint64_t *cur = c->dirty_a, *nxt = c->dirty_b;
+    for (uint32_t l = 0; l < c->cap_log2; l++) {
+      hipLaunchKernelGGL(k_cache_propagate, dim3(blocks), dim3(64), 0,
+                         ctx->stream, c->levels, c->cap_log2, l, cur, m, nxt);
+      uint64_t *t = cur;
+      cur = nxt;
+      nxt = t;
+    }
+  }
+  return m3x_registry_cache_root(ctx, c, out_root);
+}
+
+} // extern "C"

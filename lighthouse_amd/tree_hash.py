@@ -129,3 +129,58 @@ def merkleize_batch(chunk_groups, ctx=None):
     if rc != 0:
         raise RuntimeError(f"m3x_merkleize_batch rc={rc}")
     return [out.raw[32 * i : 32 * (i + 1)] for i in range(n)]
+
+
+class RegistryCache:
+    """Incremental validator-registry merkleize (SURVEY §8f.3 — the
+    milhouse cached-rehash seam, beacon_state.rs:1990-2021): the full tree
+    lives in HBM; updates rehash only dirty leaves + root paths."""
+
+    def __init__(self, ssz: bytes, n: int, ctx=None):
+        import ctypes as ct
+
+        self._ctx = ctx or _native.default_ctx()
+        self._h = ct.c_void_p()
+        rc = self._ctx._lib.m3x_registry_cache_create(
+            self._ctx.handle, ssz, n, ct.byref(self._h)
+        )
+        if rc != 0:
+            raise RuntimeError(f"m3x_registry_cache_create rc={rc}")
+
+    def close(self):
+        if self._h:
+            self._ctx._lib.m3x_registry_cache_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def root(self) -> bytes:
+        import ctypes as ct
+
+        out = ct.create_string_buffer(32)
+        rc = self._ctx._lib.m3x_registry_cache_root(
+            self._ctx.handle, self._h, out
+        )
+        if rc != 0:
+            raise RuntimeError(f"m3x_registry_cache_root rc={rc}")
+        return out.raw
+
+    def update(self, indices, records: bytes) -> bytes:
+        """apply updated/appended 121B records at `indices`; returns the
+        new List[Validator, 2^40] root."""
+        import ctypes as ct
+
+        m = len(indices)
+        assert len(records) == 121 * m
+        idx = (ct.c_uint64 * max(m, 1))(*indices)
+        out = ct.create_string_buffer(32)
+        rc = self._ctx._lib.m3x_registry_cache_update(
+            self._ctx.handle, self._h, idx, records, m, out
+        )
+        if rc != 0:
+            raise RuntimeError(f"m3x_registry_cache_update rc={rc}")
+        return out.raw

@@ -82,3 +82,47 @@ def test_basic_list_roots(ctx):
     data = bytes(rng.getrandbits(8) for _ in range(3001))
     got = th.basic_list_root(data, 3001, 1, 1 << 40, ctx=ctx)
     assert got == ssz_ref.basic_list_root(data, 3001, 1, 1 << 40)
+
+
+def test_registry_cache_incremental(ctx, oracle):
+    """incremental cache: build, mutate, append — every root
+    bit-exact vs the oracle's full rebuild of the mutated registry."""
+    import ctypes
+    import random
+
+    from lighthouse_amd import tree_hash as th
+
+    n = 10000
+    recs = [bytearray(ssz_ref.synthetic_validator_ssz(i)) for i in range(n)]
+
+    def oracle_root(count):
+        blob = b"".join(bytes(r) for r in recs[:count])
+        out = ctypes.create_string_buffer(32)
+        oracle.m3x_oracle_validator_registry_root(
+            blob, ctypes.c_uint64(count), out
+        )
+        return out.raw
+
+    cache = th.RegistryCache(b"".join(bytes(r) for r in recs), n, ctx=ctx)
+    assert cache.root() == oracle_root(n)
+
+    # mutate 200 random validators (balance + slashed flag)
+    rng = random.Random(11)
+    idxs = sorted(rng.sample(range(n), 200))
+    blob = b""
+    for i in idxs:
+        recs[i][80:88] = int(17_000_000_000 + i).to_bytes(8, "little")
+        recs[i][88] = 1
+        blob += bytes(recs[i])
+    assert cache.update(idxs, blob) == oracle_root(n)
+
+    # append 50 new validators
+    new_idx = list(range(n, n + 50))
+    for i in new_idx:
+        recs.append(bytearray(ssz_ref.synthetic_validator_ssz(i)))
+    blob = b"".join(bytes(recs[i]) for i in new_idx)
+    assert cache.update(new_idx, blob) == oracle_root(n + 50)
+
+    # empty update: root unchanged
+    assert cache.update([], b"") == oracle_root(n + 50)
+    cache.close()
