@@ -261,8 +261,8 @@ def main():
     ctx.timing_enable(True)
     bls_step()
     merkle_step()
+    ktimes = ctx.kernel_times()  # read BEFORE disable (disable resets)
     ctx.timing_enable(False)
-    ktimes = ctx.kernel_times()
 
     # C4: synthetic Deneb block import (131 sets: 128 aggregates k=512 +
     # proposal + randao + sync-agg as k=1) + full state root
