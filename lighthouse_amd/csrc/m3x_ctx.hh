@@ -21,7 +21,9 @@ struct m3x_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
   hipStream_t stream2 = nullptr; // overlap of independent kernels
+  hipStream_t stream3 = nullptr; // third pipeline stage
   hipEvent_t ev_s2 = nullptr;
+  hipEvent_t ev_pipe[24] = {};   // chunk-pipeline dependencies
   // device-resident zero-hash ladder Z[0..64] (computed on GPU at create)
   uint8_t *zeros_dev = nullptr;
   // growable scratch buffers for merkle reduction stages

@@ -295,11 +295,18 @@ int32_t m3x_ctx_create(m3x_ctx **out, int32_t device) {
   }
   if (hipStreamCreate(&ctx->stream) != hipSuccess ||
       hipStreamCreate(&ctx->stream2) != hipSuccess ||
+      hipStreamCreate(&ctx->stream3) != hipSuccess ||
       hipEventCreateWithFlags(&ctx->ev_s2, hipEventDisableTiming) !=
           hipSuccess) {
     delete ctx;
     return M3X_ERR_HIP;
   }
+  for (int i = 0; i < 24; i++)
+    if (hipEventCreateWithFlags(&ctx->ev_pipe[i], hipEventDisableTiming) !=
+        hipSuccess) {
+      delete ctx;
+      return M3X_ERR_HIP;
+    }
   if (hipMalloc(&ctx->zeros_dev, 65 * 32) != hipSuccess) {
     delete ctx;
     return M3X_ERR_NOMEM;
@@ -327,7 +334,10 @@ void m3x_ctx_destroy(m3x_ctx *ctx) {
   if (ctx->scratch_b) (void)hipFree(ctx->scratch_b);
   if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
   if (ctx->stream2) (void)hipStreamDestroy(ctx->stream2);
+  if (ctx->stream3) (void)hipStreamDestroy(ctx->stream3);
   if (ctx->ev_s2) (void)hipEventDestroy(ctx->ev_s2);
+  for (int i = 0; i < 24; i++)
+    if (ctx->ev_pipe[i]) (void)hipEventDestroy(ctx->ev_pipe[i]);
   delete ctx;
 }
 
