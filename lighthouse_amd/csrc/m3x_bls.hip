@@ -260,49 +260,28 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.gt_parts = reinterpret_cast<fp12m *>(base + off_gt);
   w.verdict = reinterpret_cast<int *>(base + off_v);
   M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
-  // Chunked 3-stream pipeline: prepare (s1), h2c (s2) and miller (s3)
-  // chunks overlap — each kernel alone fills only ~1 wave/SIMD at 64k
-  // sets, so co-residency of different stages recovers the idle issue
-  // slots. Chunk c's miller waits on events from its prepare and h2c.
-  M3X_HIP_CHECK(hipEventRecord(ctx->ev_pipe[23], ctx->stream));
-  M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream2, ctx->ev_pipe[23], 0));
-  M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream3, ctx->ev_pipe[23], 0));
-  const uint64_t NCH = (n >= 16384) ? 4 : 1;
-  uint64_t per_ch = (n + NCH - 1) / NCH;
-  for (uint64_t c = 0; c < NCH; c++) {
-    uint64_t st = c * per_ch;
-    if (st >= n) break;
-    uint64_t cnt = (st + per_ch <= n) ? per_ch : (n - st);
-    uint32_t cblocks = (uint32_t)((cnt + 63) / 64);
-    BlsWork wc = w;
-    wc.p_scaled = w.p_scaled + st;
-    wc.rsig = w.rsig + st;
-    wc.h2c = w.h2c + st;
-    wc.fparts = w.fparts + st;
-    m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
-    hipLaunchKernelGGL(k_bls_prepare, dim3(cblocks), dim3(64), 0, ctx->stream,
-                       (const uint8_t *)sigs_dev + 96 * st,
-                       (const uint8_t *)pks_dev,
-                       (const uint32_t *)offs_dev + st,
-                       (const uint64_t *)rands_dev + st, cnt, wc);
-    m3x::time_end(ctx, M3X_K_BLS_PREPARE);
-    M3X_HIP_CHECK(hipEventRecord(ctx->ev_pipe[c], ctx->stream));
-    m3x::time_begin_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
-    hipLaunchKernelGGL(k_bls_h2c, dim3(cblocks), dim3(64), 0, ctx->stream2,
-                       (const uint8_t *)msgs_dev + 32 * st, cnt, wc);
-    m3x::time_end_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
-    M3X_HIP_CHECK(hipEventRecord(ctx->ev_pipe[8 + c], ctx->stream2));
-    M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream3, ctx->ev_pipe[c], 0));
-    M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream3, ctx->ev_pipe[8 + c], 0));
-    m3x::time_begin_s(ctx, M3X_K_BLS_MILLER, ctx->stream3);
-    hipLaunchKernelGGL(k_bls_miller, dim3(cblocks), dim3(64), 0, ctx->stream3,
-                       cnt, wc);
-    m3x::time_end_s(ctx, M3X_K_BLS_MILLER, ctx->stream3);
-  }
-  // main stream continues after the last miller chunk
-  M3X_HIP_CHECK(hipEventRecord(ctx->ev_pipe[16], ctx->stream3));
-  M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream, ctx->ev_pipe[16], 0));
-  DBG_STEP(ctx, "pipeline");
+  uint32_t blocks = (uint32_t)((n + 63) / 64);
+  m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
+  hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
+                     (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
+                     (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
+                     n, w);
+  m3x::time_end(ctx, M3X_K_BLS_PREPARE);
+  DBG_STEP(ctx, "prepare");
+  // h2c is independent of prepare: run it on the second stream so the two
+  // ~1-wave/SIMD kernels co-reside (both fit at 2 waves/SIMD by VGPR count)
+  m3x::time_begin_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
+  hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream2,
+                     (const uint8_t *)msgs_dev, n, w);
+  m3x::time_end_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
+  M3X_HIP_CHECK(hipEventRecord(ctx->ev_s2, ctx->stream2));
+  M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream, ctx->ev_s2, 0));
+  DBG_STEP(ctx, "h2c");
+  m3x::time_begin(ctx, M3X_K_BLS_MILLER);
+  hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
+                     w);
+  m3x::time_end(ctx, M3X_K_BLS_MILLER);
+  DBG_STEP(ctx, "miller");
   uint32_t rblocks = (uint32_t)((n + 255) / 256);
   if (rblocks > 256) rblocks = 256;
   m3x::time_begin(ctx, M3X_K_BLS_REDUCE);
