@@ -1100,13 +1100,10 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
   if (Pa.inf || Qa.inf) return; // e(O,.) = e(.,O) = 1
   g2j T;
   g2j_from_aff(T, Qa);
+  // xi^-1 = (1+u)^-1 = (1-u)/2 — a constant, no field inversion needed
   fp2 xi_inv;
-  {
-    fp2 xi, one;
-    fp2_one(one);
-    fp2_mul_xi(xi, one);
-    fp2_inv(xi_inv, xi);
-  }
+  FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
+  fp_neg(xi_inv.c1, xi_inv.c0);
   fp12m *cur = &out, *tm = &tmp;
   fp xp = Pa.x, yp = Pa.y;
   for (int i = 62; i >= 0; i--) {
@@ -1394,13 +1391,9 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
   if (Pa.inf || Qa.inf) return;
   g2j T;
   g2j_from_aff(T, Qa);
-  fp2 xi_inv;
-  {
-    fp2 xi, one;
-    fp2_one(one);
-    fp2_mul_xi(xi, one);
-    fp2_inv(xi_inv, xi);
-  }
+  fp2 xi_inv; // constant (1-u)/2, see miller_raw
+  FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
+  fp_neg(xi_inv.c1, xi_inv.c0);
   fp xp = Pa.x, yp = Pa.y;
   for (int i = 62; i >= 0; i--) {
     f12_sqr_w(out, ws, lane);
