@@ -905,6 +905,24 @@ __device__ inline int g1_decompress(g1a &r, const uint8_t *in) {
   return 0;
 }
 
+// trusted variant for pubkey-cache bytes: range checks only, no on-curve
+// re-verification (the cache validated once at build — the same contract
+// as blst's pk_validate=false flags in verify_multiple_aggregate_signatures)
+__device__ inline int g1_from_uncomp_trusted(g1a &r, const uint8_t *in) {
+  if (in[0] & 0x40) {
+    for (int i = 0; i < 96; i++)
+      if (in[i] != (i == 0 ? 0x40 : 0)) return -1;
+    r.inf = 1;
+    fp_zero(r.x);
+    fp_zero(r.y);
+    return 0;
+  }
+  if (!fp_from_be48(r.x, in)) return -1;
+  if (!fp_from_be48(r.y, in + 48)) return -1;
+  r.inf = 0;
+  return 0;
+}
+
 __device__ inline int g1_from_uncomp(g1a &r, const uint8_t *in) {
   if (in[0] & 0x40) {
     for (int i = 0; i < 96; i++)

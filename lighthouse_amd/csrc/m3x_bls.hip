@@ -104,7 +104,7 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
   fp_zero(apk.z);
   for (uint32_t k = k0; k < k1; k++) {
     g1a pk;
-    if (g1_from_uncomp(pk, pks + 96 * (uint64_t)k) != 0) {
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k) != 0) {
       atomicOr(w.fail, 1);
       return;
     }
