@@ -192,11 +192,11 @@ def main():
         assert rc == 0, rc
         node = out32.raw
         if world > 1:
+            # NCCL has no gather primitive: use all_gather (8x32B, latency-
+            # bound; SURVEY 8e — never a ring for 32 bytes)
             t = torch.frombuffer(bytearray(node), dtype=torch.uint8).clone().cuda()
-            gathered = (
-                [torch.zeros_like(t) for _ in range(world)] if rank == 0 else None
-            )
-            dist.gather(t, gathered, dst=0)
+            gathered = [torch.zeros_like(t) for _ in range(world)]
+            dist.all_gather(gathered, t)
             if rank == 0:
                 nodes = [bytes(g.cpu().numpy().tobytes()) for g in gathered]
                 level = sub_depth

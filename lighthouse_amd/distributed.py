@@ -77,8 +77,9 @@ def registry_root_sharded(subtree_fn, hash2_fn, finalize_fn, n_leaves,
     t = torch.frombuffer(bytearray(root), dtype=torch.uint8).clone()
     if dist.get_backend(group) == "nccl":
         t = t.cuda()
-    gathered = [torch.zeros_like(t) for _ in range(world)] if rank == 0 else None
-    dist.gather(t, gathered if rank == 0 else None, dst=0, group=group)
+    # all_gather: NCCL has no gather primitive (8x32B, latency-bound)
+    gathered = [torch.zeros_like(t) for _ in range(world)]
+    dist.all_gather(gathered, t, group=group)
     if rank != 0:
         return None
     nodes = [bytes(g.cpu().numpy().tobytes()) for g in gathered]
