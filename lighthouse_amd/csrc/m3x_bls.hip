@@ -30,6 +30,9 @@ using namespace m3xb;
 namespace {
 
 struct BlsWork {
+  g1j *apk;        // [n] precomputed aggregate pubkeys (k>1 sets)
+  uint64_t *agg_idx; // [n] indices of k>1 sets (count in agg_count[0])
+  uint32_t *agg_count;
   g1a *p_scaled;   // [n] r_i * aggregate pubkey (affine)
   g2a *h2c;        // [n] hash_to_curve(msg)
   g2j *rsig;       // [n] r_i * sigma (jacobian)
@@ -79,6 +82,54 @@ __global__ __launch_bounds__(64, 1) void k_bls_pk_decompress(const uint8_t *__re
   status[i] = 0;
 }
 
+// collect the k>1 sets (their aggregation runs wave-parallel)
+__global__ void k_bls_scan_agg(const uint32_t *__restrict__ offs, uint64_t n,
+                               BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  if (offs[i + 1] - offs[i] > 1) {
+    uint32_t pos = atomicAdd(w.agg_count, 1u);
+    w.agg_idx[pos] = i;
+  }
+}
+
+// one WAVE per k>1 set: each lane partial-sums a strided slice of the
+// set's pubkeys in Jacobian form, then an LDS tree reduce (6 levels)
+__global__ __launch_bounds__(64, 1) void k_bls_aggregate_w(
+    const uint8_t *__restrict__ pks, const uint32_t *__restrict__ offs,
+    BlsWork w) {
+  __shared__ g1j lds[64];
+  if (blockIdx.x >= *w.agg_count) return;
+  uint64_t set = w.agg_idx[blockIdx.x];
+  uint32_t k0 = offs[set], k1 = offs[set + 1];
+  int lane = threadIdx.x;
+  g1j acc;
+  fp_zero(acc.x);
+  fp_zero(acc.y);
+  fp_zero(acc.z);
+  bool bad = false;
+  for (uint32_t k = k0 + lane; k < k1; k += 64) {
+    g1a pk;
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k) != 0) {
+      bad = true;
+      break;
+    }
+    g1j_add_aff(acc, acc, pk);
+  }
+  if (bad) atomicOr(w.fail, 1);
+  lds[lane] = acc;
+  __syncthreads();
+  for (int sft = 32; sft > 0; sft >>= 1) {
+    if (lane < sft) {
+      g1j t;
+      g1j_add(t, lds[lane], lds[lane + sft]);
+      lds[lane] = t;
+    }
+    __syncthreads();
+  }
+  if (lane == 0) w.apk[set] = lds[0];
+}
+
 // per-set prepare: sig decompress + subgroup check, pk aggregation,
 // r-scaling of both sides
 __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict__ sigs,
@@ -99,16 +150,15 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
     return;
   }
   g1j apk;
-  fp_zero(apk.x);
-  fp_zero(apk.y);
-  fp_zero(apk.z);
-  for (uint32_t k = k0; k < k1; k++) {
+  if (k1 - k0 == 1) {
     g1a pk;
-    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k) != 0) {
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k0) != 0) {
       atomicOr(w.fail, 1);
       return;
     }
-    g1j_add_aff(apk, apk, pk);
+    g1j_from_aff(apk, pk);
+  } else {
+    apk = w.apk[i]; // precomputed by k_bls_aggregate_w
   }
   if (g1j_is_inf(apk)) { // aggregate at infinity -> invalid
     atomicOr(w.fail, 1);
@@ -235,6 +285,9 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   BlsWork w;
   uint64_t bytes = 0;
   auto align = [](uint64_t x) { return (x + 255) & ~255ull; };
+  uint64_t off_apk = bytes; bytes += align(n * sizeof(g1j));
+  uint64_t off_aggidx = bytes; bytes += align(n * 8);
+  uint64_t off_aggcnt = bytes; bytes += 256;
   uint64_t off_p = bytes; bytes += align(n * sizeof(g1a));
   uint64_t off_h = bytes; bytes += align(n * sizeof(g2a));
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
@@ -249,6 +302,9 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                                bytes);
   if (rc != M3X_OK) return rc;
   uint8_t *base = ctx->scratch_a;
+  w.apk = reinterpret_cast<g1j *>(base + off_apk);
+  w.agg_idx = reinterpret_cast<uint64_t *>(base + off_aggidx);
+  w.agg_count = reinterpret_cast<uint32_t *>(base + off_aggcnt);
   w.p_scaled = reinterpret_cast<g1a *>(base + off_p);
   w.h2c = reinterpret_cast<g2a *>(base + off_h);
   w.rsig = reinterpret_cast<g2j *>(base + off_r);
@@ -260,8 +316,14 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.gt_parts = reinterpret_cast<fp12m *>(base + off_gt);
   w.verdict = reinterpret_cast<int *>(base + off_v);
   M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
+  M3X_HIP_CHECK(hipMemsetAsync(w.agg_count, 0, 4, ctx->stream));
   uint32_t blocks = (uint32_t)((n + 63) / 64);
   m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
+  hipLaunchKernelGGL(k_bls_scan_agg, dim3(blocks), dim3(64), 0, ctx->stream,
+                     (const uint32_t *)offs_dev, n, w);
+  hipLaunchKernelGGL(k_bls_aggregate_w, dim3((uint32_t)n), dim3(64), 0,
+                     ctx->stream, (const uint8_t *)pks_dev,
+                     (const uint32_t *)offs_dev, w);
   hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
                      (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
