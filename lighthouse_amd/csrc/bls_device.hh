@@ -1676,43 +1676,56 @@ __device__ inline void g2j_neg(g2j &r, const g2j &p) {
   r.z = p.z;
 }
 
-__device__ inline void clear_cofactor_g2(g2a &out, const g2a &p) {
-  // Budroni-Pintore (== RFC h_eff; validated by generator)
-  g2j t;
-  g2a xp_a, xxp_a;
-  g2j_mul_u64(t, p, BLS_X_ABS);
-  g2j_neg(t, t);
-  g2j_to_aff(xp_a, t); // [x]P
-  g2j_mul_u64(t, xp_a, BLS_X_ABS);
-  g2j_neg(t, t);
-  g2j_to_aff(xxp_a, t); // [x^2]P
-  g2j acc, tmp;
-  g2j_from_aff(acc, xxp_a);
-  g2j_from_aff(tmp, xp_a);
-  g2j_neg(tmp, tmp);
+// psi on Jacobian coordinates: psi(X,Y,Z) = (cx*conj(X), cy*conj(Y),
+// conj(Z)) — X'/Z'^2 = cx*conj(X/Z^2), Y'/Z'^3 = cy*conj(Y/Z^3); no
+// inversion needed.
+__device__ inline void psi_g2j(g2j &r, const g2j &p) {
+  fp2 cx, cy, t;
+  FP2_LOAD_C(cx, PSI_CX);
+  FP2_LOAD_C(cy, PSI_CY);
+  fp2_conj(t, p.x);
+  fp2_mul(r.x, t, cx);
+  fp2_conj(t, p.y);
+  fp2_mul(r.y, t, cy);
+  fp2_conj(r.z, p.z);
+}
+
+// [k]P with a Jacobian base (the few additions use the full add)
+__device__ inline void g2j_mul_u64_j(g2j &r, const g2j &base, uint64_t k) {
+  g2j acc;
+  fp2_zero(acc.x);
+  fp2_zero(acc.y);
+  fp2_zero(acc.z);
+  for (int b = 63; b >= 0; b--) {
+    g2j_dbl(acc, acc);
+    if ((k >> b) & 1) g2j_add(acc, acc, base);
+  }
+  r = acc;
+}
+
+__device__ inline void clear_cofactor_g2j(g2j &out, const g2j &p) {
+  // Budroni-Pintore (== RFC h_eff; validated by generator), fully
+  // inversion-free in Jacobian coordinates.
+  g2j xp, xxp, acc, tmp, d;
+  g2j_mul_u64_j(xp, p, BLS_X_ABS);
+  g2j_neg(xp, xp); // [x]P
+  g2j_mul_u64_j(xxp, xp, BLS_X_ABS);
+  g2j_neg(xxp, xxp); // [x^2]P
+  acc = xxp;
+  g2j_neg(tmp, xp);
   g2j_add(acc, acc, tmp);
-  g2j_from_aff(tmp, p);
-  g2j_neg(tmp, tmp);
+  g2j_neg(tmp, p);
   g2j_add(acc, acc, tmp); // [x^2-x-1]P
-  g2j d;
-  g2j_from_aff(d, xp_a);
-  g2j_from_aff(tmp, p);
-  g2j_neg(tmp, tmp);
+  d = xp;
+  g2j_neg(tmp, p);
   g2j_add(d, d, tmp);
-  g2a d_a, psi_a;
-  g2j_to_aff(d_a, d);
-  psi_g2(psi_a, d_a);
-  g2j_from_aff(tmp, psi_a);
+  psi_g2j(tmp, d);
   g2j_add(acc, acc, tmp); // + [x-1]psi(P)
-  g2j_from_aff(tmp, p);
-  g2j_dbl(tmp, tmp);
-  g2a two_a;
-  g2j_to_aff(two_a, tmp);
-  psi_g2(psi_a, two_a);
-  psi_g2(psi_a, psi_a);
-  g2j_from_aff(tmp, psi_a);
+  g2j_dbl(tmp, p);
+  psi_g2j(tmp, tmp);
+  psi_g2j(tmp, tmp);
   g2j_add(acc, acc, tmp); // + psi^2([2]P)
-  g2j_to_aff(out, acc);
+  out = acc;
 }
 
 __device__ inline void h2c_g2(g2a &r, const uint8_t *msg) {
@@ -1728,13 +1741,12 @@ __device__ inline void h2c_g2(g2a &r, const uint8_t *msg) {
   sswu_g2(q1p, u1);
   iso_map_g2(q0, q0p);
   iso_map_g2(q1, q1p);
-  g2j s, t;
+  g2j s, t, cleared;
   g2j_from_aff(s, q0);
   g2j_from_aff(t, q1);
   g2j_add(s, s, t);
-  g2a sum;
-  g2j_to_aff(sum, s);
-  clear_cofactor_g2(r, sum);
+  clear_cofactor_g2j(cleared, s);
+  g2j_to_aff(r, cleared); // single inversion, at the very end
 }
 
 // G1 generator (negated y variant computed by callers when needed)
