@@ -154,6 +154,13 @@ int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp /* n*48 */,
  * Host-side rules the caller keeps (as the reference keeps them above
  * blst): empty set list, empty (all-zero) signatures, empty key lists.
  * Returns 1 valid / 0 invalid / <0 error. */
+/* Aggregate n compressed G2 signatures into one (the TAggregateSignature
+ * add_assign surface, generic_aggregate_signature.rs:124-150; used by
+ * attestation aggregation). Invalid encodings -> nonzero return; the sum
+ * may legitimately be the point at infinity (0xc0...). */
+int32_t m3x_bls_sig_aggregate(m3x_ctx *ctx, const uint8_t *sigs /* n*96 */,
+                              uint64_t n, uint8_t out_sig[96]);
+
 int32_t m3x_bls_verify_sets(m3x_ctx *ctx, const uint8_t *msgs,
                             const uint8_t *sigs, const uint8_t *pks,
                             const uint32_t *pk_offsets, const uint64_t *rands,

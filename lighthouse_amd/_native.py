@@ -84,6 +84,11 @@ def _bind(lib):
         ),
         ("m3x_timing_enable", [ctypes.c_void_p, ctypes.c_int32]),
         (
+            "m3x_bls_sig_aggregate",
+            [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint64,
+             ctypes.c_char_p],
+        ),
+        (
             "m3x_registry_cache_create",
             [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_uint64,
              ctypes.POINTER(ctypes.c_void_p)],

@@ -121,6 +121,26 @@ class Signature:
         return self._comp if self._comp is not None else NONE_SIGNATURE
 
 
+def aggregate_signatures(sigs, ctx=None) -> "Signature":
+    """Aggregate signatures on GPU (TAggregateSignature::add_assign chain,
+    generic_aggregate_signature.rs:124-150): empty inputs are skipped (the
+    reference's empty value contributes nothing); returns the compressed
+    sum (possibly infinity)."""
+    import ctypes as ct
+
+    blobs = [s.serialize() for s in sigs if not s.is_empty()]
+    if not blobs:
+        return Signature.infinity()
+    ctx = ctx or _native.default_ctx()
+    out = ct.create_string_buffer(96)
+    rc = ctx._lib.m3x_bls_sig_aggregate(
+        ctx.handle, b"".join(blobs), len(blobs), out
+    )
+    if rc != 0:
+        raise Error(f"m3x_bls_sig_aggregate rc={rc}")
+    return Signature.from_compressed(out.raw)
+
+
 AggregateSignature = Signature
 
 

@@ -279,6 +279,57 @@ __global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
   if (lane == 0) *w.verdict = f12_is_one(sh[2]) ? 1 : 0;
 }
 
+// one wave aggregates n compressed signatures: lane-strided decompress +
+// Jacobian partial sums, LDS tree reduce, lane 0 compresses the total
+__global__ __launch_bounds__(64, 1) void k_bls_sig_aggregate_w(
+    const uint8_t *__restrict__ sigs, uint64_t n, uint8_t *__restrict__ out,
+    int *__restrict__ fail) {
+  __shared__ g2j lds[64];
+  int lane = threadIdx.x;
+  g2j acc;
+  fp2_zero(acc.x);
+  fp2_zero(acc.y);
+  fp2_zero(acc.z);
+  bool bad = false;
+  for (uint64_t i = lane; i < n; i += 64) {
+    g2a sg;
+    if (g2_decompress(sg, sigs + 96 * i) != 0) {
+      bad = true;
+      break;
+    }
+    if (!sg.inf) {
+      g2j sj;
+      g2j_from_aff(sj, sg);
+      g2j_add(acc, acc, sj);
+    }
+  }
+  if (bad) atomicOr(fail, 1);
+  lds[lane] = acc;
+  __syncthreads();
+  for (int sft = 32; sft > 0; sft >>= 1) {
+    if (lane < sft) {
+      g2j t;
+      g2j_add(t, lds[lane], lds[lane + sft]);
+      lds[lane] = t;
+    }
+    __syncthreads();
+  }
+  if (lane == 0 && !*fail) {
+    g2a a;
+    g2j_to_aff(a, lds[0]);
+    // compress (ZCash): c1 || c0 big-endian, flags in byte 0
+    if (a.inf) {
+      for (int i = 0; i < 96; i++) out[i] = 0;
+      out[0] = 0xC0;
+    } else {
+      fp_to_be48(a.x.c1, out);
+      fp_to_be48(a.x.c0, out + 48);
+      out[0] |= 0x80;
+      if (fp2_gt_half_lex(a.y)) out[0] |= 0x20;
+    }
+  }
+}
+
 int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                const void *pks_dev, const void *offs_dev,
                const void *rands_dev, uint64_t n, int32_t *out) {
@@ -394,6 +445,32 @@ int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp, uint64_t n,
   (void)hipFree(unc_d);
   (void)hipFree(st_d);
   return e == hipSuccess ? M3X_OK : M3X_ERR_HIP;
+}
+
+int32_t m3x_bls_sig_aggregate(m3x_ctx *ctx, const uint8_t *sigs, uint64_t n,
+                              uint8_t out_sig[96]) {
+  if (!ctx || n == 0) return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *sigs_d, *out_d;
+  int *fail_d;
+  M3X_HIP_CHECK(hipMalloc(&sigs_d, n * 96));
+  M3X_HIP_CHECK(hipMalloc(&out_d, 96));
+  M3X_HIP_CHECK(hipMalloc(&fail_d, 4));
+  hipError_t e1 = hipMemcpyAsync(sigs_d, sigs, n * 96,
+                                 hipMemcpyHostToDevice, ctx->stream);
+  (void)hipMemsetAsync(fail_d, 0, 4, ctx->stream);
+  hipLaunchKernelGGL(k_bls_sig_aggregate_w, dim3(1), dim3(64), 0, ctx->stream,
+                     sigs_d, n, out_d, fail_d);
+  int fail = 1;
+  (void)hipMemcpyAsync(out_sig, out_d, 96, hipMemcpyDeviceToHost, ctx->stream);
+  (void)hipMemcpyAsync(&fail, fail_d, 4, hipMemcpyDeviceToHost, ctx->stream);
+  hipError_t e2 = hipStreamSynchronize(ctx->stream);
+  (void)hipFree(sigs_d);
+  (void)hipFree(out_d);
+  (void)hipFree(fail_d);
+  if (e1 != hipSuccess || e2 != hipSuccess) return M3X_ERR_HIP;
+  return fail ? M3X_ERR_ARG : M3X_OK;
 }
 
 int32_t m3x_bls_verify_sets_dev(m3x_ctx *ctx, const void *msgs_dev,

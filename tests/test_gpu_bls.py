@@ -244,3 +244,34 @@ def test_aggregate_pubkey_at_infinity_rejected(ctx, oracle):
         )
         == 0
     )
+
+
+def test_signature_aggregation_matches_aggregate_secret(ctx, oracle):
+    """GPU aggregation of k signatures over one message must equal the
+    signature under the SUM of the secret keys (bilinearity of signing) —
+    byte-exact, and verify under fast_aggregate_verify."""
+    import ctypes as ct
+
+    from lighthouse_amd import bls
+
+    ORDER = 0x73EDA753299D7D483339D80809A1D80553BDA402FFFE5BFEFFFFFFFF00000001
+    msg = b"\x5a" * 32
+    sks, sigs, pks = [], [], []
+    skb = ct.create_string_buffer(32)
+    sgb = ct.create_string_buffer(96)
+    for i in range(7):
+        oracle.m3x_oracle_bls_keygen(ct.c_uint64(i), skb)
+        sks.append(int.from_bytes(skb.raw, "big"))
+        assert oracle.m3x_oracle_bls_sign(skb.raw, msg, sgb) == 0
+        sigs.append(bls.Signature.from_compressed(sgb.raw))
+        pks.append(
+            bls.PublicKey.from_uncompressed(
+                bytes.fromhex(FIXTURES["interop"][i]["pk_uncompressed_hex"])
+            )
+        )
+    agg = bls.aggregate_signatures(sigs, ctx=ctx)
+    sk_sum = (sum(sks) % ORDER).to_bytes(32, "big")
+    want = ct.create_string_buffer(96)
+    assert oracle.m3x_oracle_bls_sign(sk_sum, msg, want) == 0
+    assert agg.serialize() == want.raw
+    assert bls.fast_aggregate_verify(agg, msg, pks, ctx=ctx) is True
