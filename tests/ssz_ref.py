@@ -189,3 +189,15 @@ def beacon_state_root_ref(st):
                             len(st["historical_summaries"]), 1 << 24))
     assert len(f) == 28
     return _container(f)
+
+
+def signing_root_ref(object_root: bytes, domain: bytes) -> bytes:
+    """signing_data.rs:22-31: hash_tree_root(SigningData{object_root,domain})"""
+    return merkleize([object_root, domain], 1)
+
+
+def attestation_data_root_ref(slot, index, bbr, src, tgt) -> bytes:
+    ck = lambda e, r: merkleize([_c(_u64le(e)), r], 1)
+    return merkleize(
+        [_c(_u64le(slot)), _c(_u64le(index)), bbr, ck(*src), ck(*tgt)], 3
+    )

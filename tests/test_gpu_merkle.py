@@ -126,3 +126,28 @@ def test_registry_cache_incremental(ctx, oracle):
     # empty update: root unchanged
     assert cache.update([], b"") == oracle_root(n + 50)
     cache.close()
+
+
+def test_batched_signing_roots(ctx):
+    """batched signing-root construction vs the hashlib
+    restatement (signing_data.rs:22-31, AttestationData container)."""
+    import hashlib
+
+    from lighthouse_amd import signing
+
+    domain = hashlib.sha256(b"domain").digest()
+    atts = [
+        (
+            1000 + i,
+            i % 64,
+            hashlib.sha256(b"bbr%d" % i).digest(),
+            (31, hashlib.sha256(b"src%d" % i).digest()),
+            (32, hashlib.sha256(b"tgt%d" % i).digest()),
+        )
+        for i in range(200)
+    ]
+    got = signing.attestation_signing_roots(atts, domain, ctx=ctx)
+    for i, a in enumerate(atts):
+        obj = ssz_ref.attestation_data_root_ref(*a)
+        want = ssz_ref.signing_root_ref(obj, domain)
+        assert got[i] == want, i
