@@ -406,8 +406,6 @@ def main():
                 "kernel": "k_bls_miller",
                 "basis": "analytic op count (DESIGN.md Roofline); VALU workload per SURVEY 8d — not HBM/MFMA-bound",
             }
-        leaves_ms, leaves_n = ktimes.get("leaves", (0.0, 0))
-        reduce_ms, reduce_n = ktimes.get("reduce", (0.0, 0))
         full_nodes = bs.node_hash_count(N_VALIDATORS)
         sha_hps = full_nodes / merkle_only if merkle_only > 0 else None
         line = {
