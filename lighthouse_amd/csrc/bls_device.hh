@@ -794,6 +794,78 @@ __device__ inline void g2j_add(g2j &r, const g2j &p, const g2j &q) {
   r.z = z3;
 }
 
+// mixed add (Jacobian += affine), the G2 mirror of g1j_add_aff
+__device__ inline void g2j_add_aff(g2j &r, const g2j &p, const g2a &q) {
+  if (q.inf) {
+    r = p;
+    return;
+  }
+  if (g2j_is_inf(p)) {
+    g2j_from_aff(r, q);
+    return;
+  }
+  fp2 z1z1, u2, s2, t;
+  fp2_sqr(z1z1, p.z);
+  fp2_mul(u2, q.x, z1z1);
+  fp2_mul(t, p.z, z1z1);
+  fp2_mul(s2, q.y, t);
+  if (fp2_eq(u2, p.x)) {
+    if (fp2_eq(s2, p.y)) {
+      g2j_dbl(r, p);
+      return;
+    }
+    fp2_zero(r.x);
+    fp2_zero(r.y);
+    fp2_zero(r.z);
+    return;
+  }
+  fp2 h, hh, i, j, rr, v, x3, y3, z3;
+  fp2_sub(h, u2, p.x);
+  fp2_sqr(hh, h);
+  fp2_dbl(i, hh);
+  fp2_dbl(i, i);
+  fp2_mul(j, h, i);
+  fp2_sub(rr, s2, p.y);
+  fp2_dbl(rr, rr);
+  fp2_mul(v, p.x, i);
+  fp2_sqr(x3, rr);
+  fp2_sub(x3, x3, j);
+  fp2_sub(x3, x3, v);
+  fp2_sub(x3, x3, v);
+  fp2_sub(y3, v, x3);
+  fp2_mul(y3, rr, y3);
+  fp2_mul(t, p.y, j);
+  fp2_dbl(t, t);
+  fp2_sub(y3, y3, t);
+  fp2_add(z3, p.z, h);
+  fp2_sqr(z3, z3);
+  fp2_sub(z3, z3, z1z1);
+  fp2_sub(z3, z3, hh);
+  r.x = x3;
+  r.y = y3;
+  r.z = z3;
+}
+
+// [k1]P and [k2]P with ONE shared doubling chain of the base (LSB-first
+// double-and-add) — the subgroup-check [|x|]sigma and the batch [r]sigma
+// share sigma's doublings.
+__device__ inline void g2j_mul2_u64(g2j &r1, g2j &r2, const g2a &p,
+                                    uint64_t k1, uint64_t k2) {
+  g2j base, a1, a2;
+  g2j_from_aff(base, p);
+  fp2_zero(a1.x);
+  fp2_zero(a1.y);
+  fp2_zero(a1.z);
+  a2 = a1;
+  for (int b = 0; b < 64; b++) {
+    if ((k1 >> b) & 1) g2j_add(a1, a1, base);
+    if ((k2 >> b) & 1) g2j_add(a2, a2, base);
+    if (b < 63 && ((k1 | k2) >> (b + 1)) != 0) g2j_dbl(base, base);
+  }
+  r1 = a1;
+  r2 = a2;
+}
+
 __device__ inline void g2j_mul_be(g2j &r, const g2a &p, const uint8_t *be,
                                   int nbytes) {
   g2j acc, base;

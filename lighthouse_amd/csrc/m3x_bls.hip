@@ -140,7 +140,7 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   g2a sig;
-  if (g2_decompress(sig, sigs + 96 * i) != 0 || !g2_in_subgroup(sig)) {
+  if (g2_decompress(sig, sigs + 96 * i) != 0) {
     atomicOr(w.fail, 1);
     return;
   }
@@ -173,11 +173,35 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
   g1j_mul_be(rp, apk_a, rbe, 8);
   g1j_to_aff(w.p_scaled[i], rp);
   if (sig.inf) {
+    // infinity is a valid subgroup element; contributes nothing
     fp2_zero(w.rsig[i].x);
     fp2_zero(w.rsig[i].y);
     fp2_zero(w.rsig[i].z);
   } else {
-    g2j_mul_be(w.rsig[i], sig, rbe, 8);
+    // [r]sigma and the psi subgroup check's [|x|]sigma share sigma's
+    // doubling chain (blst.rs:73-77 deferred subgroup check)
+    g2j rsig_j, xsig_j;
+    g2j_mul2_u64(rsig_j, xsig_j, sig, rands[i], BLS_X_ABS);
+    w.rsig[i] = rsig_j;
+    // psi(sigma) must equal -[|x|]sigma (x < 0): cross-multiplied compare
+    g2a ps;
+    psi_g2(ps, sig);
+    bool ok;
+    if (g2j_is_inf(xsig_j)) {
+      ok = false;
+    } else {
+      fp2 z2, z3, lx, ly, ny;
+      fp2_sqr(z2, xsig_j.z);
+      fp2_mul(z3, z2, xsig_j.z);
+      fp2_mul(lx, ps.x, z2);
+      fp2_neg(ny, xsig_j.y);
+      fp2_mul(ly, ps.y, z3);
+      ok = fp2_eq(lx, xsig_j.x) && fp2_eq(ly, ny);
+    }
+    if (!ok) {
+      atomicOr(w.fail, 1);
+      return;
+    }
   }
 }
 
