@@ -1074,68 +1074,99 @@ __device__ __forceinline__ void f12_set(fp12m &a, int k, const fp2 &c) {
   a.s[2 * k + 1] = c.c1;
 }
 
-// r = a * b with register accumulators; r must NOT alias a or b (the
-// non-aliasing form avoids the scratch-resident lo/hi temporaries that
-// dominated the first-cut kernel at low occupancy).
-__device__ inline void f12_mul_nn(fp12m &r, const fp12m &a, const fp12m &b) {
-  for (int k = 0; k < 6; k++) {
-    fp2 acc, hacc;
-    fp2_zero(acc);
-    fp2_zero(hacc);
-    for (int i = 0; i <= k; i++) {
-      fp2 ai, bj, t;
-      f12_get(a, i, ai);
-      f12_get(b, k - i, bj);
-      fp2_mul(t, ai, bj);
-      fp2_add(acc, acc, t);
+// Fp6 helpers over the packed fp12m layout: the tower view is
+// A = (c0,c2,c4), B = (c1,c3,c5) with f = A + w*B, w^2 = v,
+// Fp6 = Fp2[v]/(v^3 - xi). Index mapping validated against the schoolbook
+// degree-6 multiply in Python (gen_bls_fixtures arithmetic).
+struct fp6 {
+  fp2 c[3];
+};
+
+__device__ __forceinline__ void f6_add(fp6 &r, const fp6 &a, const fp6 &b) {
+  fp2_add(r.c[0], a.c[0], b.c[0]);
+  fp2_add(r.c[1], a.c[1], b.c[1]);
+  fp2_add(r.c[2], a.c[2], b.c[2]);
+}
+__device__ __forceinline__ void f6_sub(fp6 &r, const fp6 &a, const fp6 &b) {
+  fp2_sub(r.c[0], a.c[0], b.c[0]);
+  fp2_sub(r.c[1], a.c[1], b.c[1]);
+  fp2_sub(r.c[2], a.c[2], b.c[2]);
+}
+__device__ __forceinline__ void f6_mul_v(fp6 &r, const fp6 &a) {
+  fp2 t;
+  fp2_mul_xi(t, a.c[2]);
+  r.c[2] = a.c[1];
+  r.c[1] = a.c[0];
+  r.c[0] = t;
+}
+// schoolbook Fp6 multiply (9 fp2 muls) + v^3 = xi reduction
+__device__ __forceinline__ void f6_mul(fp6 &r, const fp6 &a, const fp6 &b) {
+  fp2 acc[5], t;
+  for (int i = 0; i < 5; i++) fp2_zero(acc[i]);
+#pragma unroll
+  for (int i = 0; i < 3; i++)
+#pragma unroll
+    for (int j = 0; j < 3; j++) {
+      fp2_mul(t, a.c[i], b.c[j]);
+      fp2_add(acc[i + j], acc[i + j], t);
     }
-    for (int i = k + 1; i < 6; i++) {
-      fp2 ai, bj, t;
-      f12_get(a, i, ai);
-      f12_get(b, k + 6 - i, bj);
-      fp2_mul(t, ai, bj);
-      fp2_add(hacc, hacc, t);
-    }
-    fp2_mul_xi(hacc, hacc);
-    fp2_add(acc, acc, hacc);
-    f12_set(r, k, acc);
-  }
+  fp2_mul_xi(t, acc[3]);
+  fp2_add(r.c[0], acc[0], t);
+  fp2_mul_xi(t, acc[4]);
+  fp2_add(r.c[1], acc[1], t);
+  r.c[2] = acc[2];
 }
 
-// specialized squaring: out_k = 2*sum_{i<j, i+j=k or k+6} a_i a_j
-// (+ a_{k/2}^2 for even k) — 21 fp2 muls instead of 36. r must not alias a.
+__device__ __forceinline__ void f12_split(const fp12m &f, fp6 &A, fp6 &B) {
+  f12_get(f, 0, A.c[0]);
+  f12_get(f, 2, A.c[1]);
+  f12_get(f, 4, A.c[2]);
+  f12_get(f, 1, B.c[0]);
+  f12_get(f, 3, B.c[1]);
+  f12_get(f, 5, B.c[2]);
+}
+__device__ __forceinline__ void f12_join(fp12m &f, const fp6 &A,
+                                         const fp6 &B) {
+  f12_set(f, 0, A.c[0]);
+  f12_set(f, 2, A.c[1]);
+  f12_set(f, 4, A.c[2]);
+  f12_set(f, 1, B.c[0]);
+  f12_set(f, 3, B.c[1]);
+  f12_set(f, 5, B.c[2]);
+}
+
+// r = a * b via the quadratic-over-cubic tower (3 Fp6 muls = 18 fp2 muls
+// vs 36 schoolbook); r may alias a or b (inputs are split out first).
+__device__ inline void f12_mul_nn(fp12m &r, const fp12m &a, const fp12m &b) {
+  fp6 A1, B1, A2, B2, aa, bb, s1, s2, cross, even, t;
+  f12_split(a, A1, B1);
+  f12_split(b, A2, B2);
+  f6_mul(aa, A1, A2);
+  f6_mul(bb, B1, B2);
+  f6_add(s1, A1, B1);
+  f6_add(s2, A2, B2);
+  f6_mul(cross, s1, s2);
+  f6_sub(cross, cross, aa);
+  f6_sub(cross, cross, bb); // A1B2 + A2B1
+  f6_mul_v(t, bb);
+  f6_add(even, aa, t); // A1A2 + v B1B2
+  f12_join(r, even, cross);
+}
+
+// r = a^2: complex squaring, 2 Fp6 muls = 12 fp2 muls; alias-safe.
 __device__ inline void f12_sqr_nn(fp12m &r, const fp12m &a) {
-  fp2 ac[6];
-  for (int i = 0; i < 6; i++) f12_get(a, i, ac[i]);
-  for (int k = 0; k < 6; k++) {
-    fp2 acc, hacc, t;
-    fp2_zero(acc);
-    fp2_zero(hacc);
-    // low part i+j == k
-    for (int i = 0; 2 * i < k; i++) {
-      fp2_mul(t, ac[i], ac[k - i]);
-      fp2_add(acc, acc, t);
-    }
-    fp2_dbl(acc, acc);
-    if (k % 2 == 0) {
-      fp2_sqr(t, ac[k / 2]);
-      fp2_add(acc, acc, t);
-    }
-    // high part i+j == k+6 (i from k+1..5, i < j)
-    int kk = k + 6;
-    for (int i = k + 1; 2 * i < kk; i++) {
-      fp2_mul(t, ac[i], ac[kk - i]);
-      fp2_add(hacc, hacc, t);
-    }
-    fp2_dbl(hacc, hacc);
-    if (kk % 2 == 0 && kk / 2 < 6) {
-      fp2_sqr(t, ac[kk / 2]);
-      fp2_add(hacc, hacc, t);
-    }
-    fp2_mul_xi(hacc, hacc);
-    fp2_add(acc, acc, hacc);
-    f12_set(r, k, acc);
-  }
+  fp6 A, B, m1, t, u, even, odd;
+  f12_split(a, A, B);
+  f6_mul(m1, A, B);
+  f6_add(t, A, B);
+  f6_mul_v(u, B);
+  f6_add(u, A, u);
+  f6_mul(t, t, u); // (A+B)(A+vB)
+  f6_sub(t, t, m1);
+  f6_mul_v(u, m1);
+  f6_sub(even, t, u); // A^2 + v B^2
+  f6_add(odd, m1, m1); // 2AB
+  f12_join(r, even, odd);
 }
 
 // r = f * (a0 + a3 w^3 + a5 w^5); r must not alias f (sparse: 18 fp2 muls)
