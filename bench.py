@@ -331,6 +331,20 @@ def main():
         incr_ms = (time.time() - tb) * 1e3
         cache.close()
 
+    # swap-or-not shuffle: 1M indices, 90 rounds
+    shuffle_ms = None
+    if rank == 0 and world == 1:
+        import numpy as _np
+
+        idx_dev = ctx.upload(_np.arange(1 << 20, dtype=_np.uint32).tobytes())
+        seed32 = hashlib.sha256(b"bench-shuffle").digest()
+        lib.m3x_shuffle_list_dev(ctx.handle, idx_dev, 1 << 20, 90, seed32, 0)
+        torch.cuda.synchronize()
+        tb = time.time()
+        lib.m3x_shuffle_list_dev(ctx.handle, idx_dev, 1 << 20, 90, seed32, 0)
+        torch.cuda.synchronize()
+        shuffle_ms = (time.time() - tb) * 1e3
+
     # split timing: one more pass of each, timed separately (for extras)
     torch.cuda.synchronize()
     tb = time.time()
@@ -438,6 +452,7 @@ def main():
                 },
                 "c4_block_import_ms": c4_ms,
                 "registry_incremental_update_2048_ms": incr_ms,
+                "shuffle_1m_90rounds_ms": shuffle_ms,
                 "cpu_sha_baseline": cpu_sha,
             },
             "roofline": roofline,
