@@ -386,6 +386,17 @@ def main():
             "kind": "port",
             "sample": f"{len(idx)} sets (75% k=1, 25% k=512), {tcpu:.1f}s, OpenMP x{cores}",
         }
+        # shuffle CPU baseline (oracle, single list)
+        import numpy as _np
+
+        idx = _np.arange(1 << 20, dtype=_np.uint32)
+        arr = (ctypes.c_uint32 * (1 << 20)).from_buffer_copy(idx.tobytes())
+        seed32 = hashlib.sha256(b"bench-shuffle").digest()
+        tb = time.time()
+        oracle.m3x_oracle_shuffle_list(
+            arr, ctypes.c_uint64(1 << 20), ctypes.c_uint8(90), seed32, 0
+        )
+        cpu_shuffle_ms = (time.time() - tb) * 1e3
         # SHA: full registry once
         tb = time.time()
         want = ctypes.create_string_buffer(32)
@@ -397,6 +408,7 @@ def main():
             "node_hashes_per_sec": REGISTRY_NODE_HASHES / tsha,
             "state_root_s": tsha,
             "cores": cores,
+            "shuffle_1m_90rounds_ms_1core": cpu_shuffle_ms,
         }
 
     # ---------------- report ----------------
