@@ -268,21 +268,6 @@ __device__ inline bool fp_sqrt(fp &r, const fp &a) {
   return true;
 }
 
-// Legendre symbol: a^((p-1)/2) == 1 (a != 0); p = 3 mod 4
-__device__ inline bool fp_is_square(const fp &a) {
-  if (fp_is_zero(a)) return true;
-  uint64_t t[6] = {BLS_P[0] - 1, BLS_P[1], BLS_P[2],
-                   BLS_P[3], BLS_P[4], BLS_P[5]};
-  uint64_t e[6];
-#pragma unroll
-  for (int i = 0; i < 6; i++)
-    e[i] = (t[i] >> 1) | (i < 5 ? (t[i + 1] << 63) : 0);
-  fp r, one;
-  fp_pow_limbs(r, a, e, 6);
-  fp_one(one);
-  return fp_eq(r, one);
-}
-
 // standard-form compare against (p-1)/2 ("lexicographically largest")
 __device__ inline bool fp_gt_half(const fp &a) {
   uint64_t s[6];
@@ -465,16 +450,6 @@ __device__ inline bool fp2_sqrt(fp2 &r, const fp2 &a) {
   if (!fp2_eq(sq, a)) return false;
   r = cand;
   return true;
-}
-
-// a is a square in Fp2 iff its norm a0^2+a1^2 is a square in Fp
-// (a^((p^2-1)/2) = norm(a)^((p-1)/2))
-__device__ inline bool fp2_is_square(const fp2 &a) {
-  fp n, t;
-  fp_sqr(n, a.c0);
-  fp_sqr(t, a.c1);
-  fp_add(n, n, t);
-  return fp_is_square(n);
 }
 
 __device__ inline bool fp2_gt_half_lex(const fp2 &y) {
@@ -1726,10 +1701,7 @@ __device__ inline void sswu_g2(g2a &out, const fp2 &u) {
   fp2_mul(t, A, x1);
   fp2_add(gx1, gx1, t);
   fp2_add(gx1, gx1, B);
-  // branch decided upfront with one Legendre test (avoids the wasted
-  // failed-sqrt chain on the ~50% of inputs where gx1 is non-square)
-  if (fp2_is_square(gx1)) {
-    fp2_sqrt(y1, gx1); // succeeds
+  if (fp2_sqrt(y1, gx1)) {
     x = x1;
     y = y1;
   } else {
