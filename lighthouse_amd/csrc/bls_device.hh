@@ -1504,29 +1504,76 @@ __device__ inline void f12_pow_xabs_w(fp12m &r, const fp12m &a, f12w_ws &ws,
   }
 }
 
-// out = miller(P, Q) (conjugated); in-place cooperative version
+// cooperative miller workspace: the 68 per-step line-coefficient records
+// (63 doublings + 5 additions for |x| = 0xd201000000010000)
+struct miller_ws {
+  fp2 T[68][3];    // T (Jacobian) BEFORE each step
+  fp2 coef[68][3]; // line (a0, a3, a5) per step
+};
+
+// out = miller(P, Q) (conjugated), one wave. Three phases: (1) the serial
+// T-chain (lane 0) storing pre-step points, (2) line coefficients for all
+// steps computed in PARALLEL across lanes, (3) the f-chain with
+// cooperative fp12 ops consuming the stored coefficients.
 __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
-                                f12w_ws &ws, int lane) {
+                                f12w_ws &ws, miller_ws &mws, int lane) {
   if (lane == 0) f12_one(out);
   f12w_sync();
   if (Pa.inf || Qa.inf) return;
-  g2j T;
-  g2j_from_aff(T, Qa);
-  fp2 xi_inv; // constant (1-u)/2, see miller_raw
+  fp2 xi_inv; // constant (1-u)/2
   FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
   fp_neg(xi_inv.c1, xi_inv.c0);
-  fp xp = Pa.x, yp = Pa.y;
-  for (int i = 62; i >= 0; i--) {
-    f12_sqr_w(out, ws, lane);
-    {
-      fp2 X2, Y2, Z2, Z3, a0, a3, a5, t, t2;
+  // ---- phase 1: serial point chain (lane 0 writes pre-step T) ----
+  if (lane == 0) {
+    g2j T;
+    g2j_from_aff(T, Qa);
+    int idx = 0;
+    for (int i = 62; i >= 0; i--) {
+      mws.T[idx][0] = T.x;
+      mws.T[idx][1] = T.y;
+      mws.T[idx][2] = T.z;
+      idx++;
+      g2j_dbl(T, T);
+      if ((BLS_X_ABS >> i) & 1) {
+        mws.T[idx][0] = T.x;
+        mws.T[idx][1] = T.y;
+        mws.T[idx][2] = T.z;
+        idx++;
+        g2j qj;
+        g2j_from_aff(qj, Qa);
+        g2j_add(T, T, qj);
+      }
+    }
+  }
+  f12w_sync();
+  // ---- phase 2: one lane per step computes its line coefficients ----
+  if (lane < 68) {
+    // derive this record's kind (doubling or addition) from the fixed
+    // step pattern of |x|
+    int idx = 0;
+    int my_kind = -1;
+    for (int i = 62; i >= 0 && my_kind < 0; i--) {
+      if (idx == lane) my_kind = 0; // doubling record
+      idx++;
+      if ((BLS_X_ABS >> i) & 1) {
+        if (my_kind < 0 && idx == lane) my_kind = 1; // addition record
+        idx++;
+      }
+    }
+    g2j T;
+    T.x = mws.T[lane][0];
+    T.y = mws.T[lane][1];
+    T.z = mws.T[lane][2];
+    fp2 a0, a3, a5, t, t2;
+    if (my_kind == 0) {
+      fp2 X2, Y2, Z2, Z3;
       fp2_sqr(X2, T.x);
       fp2_sqr(Y2, T.y);
       fp2_sqr(Z2, T.z);
       fp2_mul(Z3, Z2, T.z);
       fp2_mul(t, T.y, Z3);
       fp2_dbl(t, t);
-      fp2_mul_fp(a0, t, yp);
+      fp2_mul_fp(a0, t, Pa.y);
       fp2_mul(t, X2, T.x);
       fp2_mul_small(t, t, 3);
       fp2_dbl(t2, Y2);
@@ -1534,14 +1581,11 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
       fp2_mul(a3, t, xi_inv);
       fp2_mul(t, X2, Z2);
       fp2_mul_small(t, t, 3);
-      fp2_mul_fp(t, t, xp);
+      fp2_mul_fp(t, t, Pa.x);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
-      f12_line_w(out, a0, a3, a5, ws, lane);
-      g2j_dbl(T, T);
-    }
-    if ((BLS_X_ABS >> i) & 1) {
-      fp2 Z2, Z3, H, M, a0, a3, a5, t, t2;
+    } else {
+      fp2 Z2, Z3, H, M;
       fp2_sqr(Z2, T.z);
       fp2_mul(Z3, Z2, T.z);
       fp2_mul(t, Qa.x, Z2);
@@ -1549,19 +1593,34 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
       fp2_mul(t, Qa.y, Z3);
       fp2_sub(M, T.y, t);
       fp2_mul(t, Z3, H);
-      fp2_mul_fp(a0, t, yp);
+      fp2_mul_fp(a0, t, Pa.y);
       fp2_mul(t, M, T.x);
       fp2_mul(t2, T.y, H);
       fp2_sub(t, t, t2);
       fp2_mul(a3, t, xi_inv);
       fp2_mul(t, M, Z2);
-      fp2_mul_fp(t, t, xp);
+      fp2_mul_fp(t, t, Pa.x);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
-      f12_line_w(out, a0, a3, a5, ws, lane);
-      g2j qj;
-      g2j_from_aff(qj, Qa);
-      g2j_add(T, T, qj);
+    }
+    mws.coef[lane][0] = a0;
+    mws.coef[lane][1] = a3;
+    mws.coef[lane][2] = a5;
+  }
+  f12w_sync();
+  // ---- phase 3: the f-chain over stored coefficients ----
+  {
+    int idx = 0;
+    for (int i = 62; i >= 0; i--) {
+      f12_sqr_w(out, ws, lane);
+      f12_line_w(out, mws.coef[idx][0], mws.coef[idx][1], mws.coef[idx][2],
+                 ws, lane);
+      idx++;
+      if ((BLS_X_ABS >> i) & 1) {
+        f12_line_w(out, mws.coef[idx][0], mws.coef[idx][1], mws.coef[idx][2],
+                   ws, lane);
+        idx++;
+      }
     }
   }
   f12_conj6_w(out, lane);

@@ -285,6 +285,7 @@ __global__ __launch_bounds__(256) void k_bls_reduce_sig(
 __global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
   __shared__ fp12m sh[7];
   __shared__ f12w_ws ws;
+  __shared__ miller_ws mws;
   int lane = threadIdx.x;
   if (*w.fail) {
     if (lane == 0) *w.verdict = 0;
@@ -297,7 +298,7 @@ __global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
   fp_neg(ng1.y, ng1.y);
   if (lane == 0) f12_copy(sh[0], w.gt_parts[0]);
   f12w_sync();
-  miller_w(sh[1], ng1, sig_a, ws, lane);
+  miller_w(sh[1], ng1, sig_a, ws, mws, lane);
   f12_mul_w(sh[1], sh[0], sh[1], ws, lane); // f_total
   final_exp_w(sh[2], sh[1], &sh[3], ws, lane);
   if (lane == 0) *w.verdict = f12_is_one(sh[2]) ? 1 : 0;
