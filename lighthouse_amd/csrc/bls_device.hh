@@ -1546,24 +1546,25 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
     }
   }
   f12w_sync();
-  // ---- phase 2: one lane per step computes its line coefficients ----
-  if (lane < 68) {
+  // ---- phase 2: lanes compute the 68 line-coefficient records in
+  // parallel (68 > 64 lanes: strided, lanes 0-3 take a second record) ----
+  for (int rec = lane; rec < 68; rec += 64) {
     // derive this record's kind (doubling or addition) from the fixed
     // step pattern of |x|
     int idx = 0;
     int my_kind = -1;
     for (int i = 62; i >= 0 && my_kind < 0; i--) {
-      if (idx == lane) my_kind = 0; // doubling record
+      if (idx == rec) my_kind = 0; // doubling record
       idx++;
       if ((BLS_X_ABS >> i) & 1) {
-        if (my_kind < 0 && idx == lane) my_kind = 1; // addition record
+        if (my_kind < 0 && idx == rec) my_kind = 1; // addition record
         idx++;
       }
     }
     g2j T;
-    T.x = mws.T[lane][0];
-    T.y = mws.T[lane][1];
-    T.z = mws.T[lane][2];
+    T.x = mws.T[rec][0];
+    T.y = mws.T[rec][1];
+    T.z = mws.T[rec][2];
     fp2 a0, a3, a5, t, t2;
     if (my_kind == 0) {
       fp2 X2, Y2, Z2, Z3;
@@ -1603,9 +1604,9 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
     }
-    mws.coef[lane][0] = a0;
-    mws.coef[lane][1] = a3;
-    mws.coef[lane][2] = a5;
+    mws.coef[rec][0] = a0;
+    mws.coef[rec][1] = a3;
+    mws.coef[rec][2] = a5;
   }
   f12w_sync();
   // ---- phase 3: the f-chain over stored coefficients ----
