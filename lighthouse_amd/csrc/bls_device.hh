@@ -643,6 +643,24 @@ __device__ inline void g1j_add_aff(g1j &r, const g1j &p, const g1a &q) {
   r.z = z3;
 }
 
+// [k]P with a Jacobian base (full adds; used where the base is already
+// Jacobian, e.g. the wave-aggregated pubkey sum)
+__device__ inline void g1j_mul_be_j(g1j &r, const g1j &base,
+                                    const uint8_t *be, int nbytes) {
+  g1j acc;
+  fp_zero(acc.x);
+  fp_zero(acc.y);
+  fp_zero(acc.z);
+  for (int i = 0; i < nbytes; i++) {
+    uint8_t byte = be[i];
+    for (int b = 7; b >= 0; b--) {
+      g1j_dbl(acc, acc);
+      if ((byte >> b) & 1) g1j_add(acc, acc, base);
+    }
+  }
+  r = acc;
+}
+
 __device__ inline void g1j_mul_be(g1j &r, const g1a &p, const uint8_t *be,
                                   int nbytes) {
   g1j acc, base;
@@ -1215,18 +1233,26 @@ __device__ inline bool f12_is_one(const fp12m &a) {
 // out = miller(P, Q) (already conjugated for x<0); tmp is scratch; out and
 // tmp must be distinct. Inversion-free Jacobian loop on the twist with
 // sparse lines (formulas validated against the Python reference).
-__device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
+__device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1j &Pj,
                                   const g2a &Qa) {
   f12_one(out);
-  if (Pa.inf || Qa.inf) return; // e(O,.) = e(.,O) = 1
+  if (fp_is_zero(Pj.z) || Qa.inf) return; // e(O,.) = e(.,O) = 1
   g2j T;
   g2j_from_aff(T, Qa);
-  // xi^-1 = (1+u)^-1 = (1-u)/2 — a constant, no field inversion needed
-  fp2 xi_inv;
+  // P stays JACOBIAN: lines scaled by the Fp-subfield factor Zp^3
+  // (final-exp invariant; validated in Python): a0 uses Yp directly,
+  // a3's xi^-1 absorbs Zp^3, a5 uses Xp*Zp — identical per-step cost and
+  // no P to-affine inversion. xi^-1 = (1-u)/2 is a constant.
+  fp2 xi_inv, xi_inv_zp3;
   FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
   fp_neg(xi_inv.c1, xi_inv.c0);
+  fp zp2, zp3, xpzp;
+  fp_sqr(zp2, Pj.z);
+  fp_mul(zp3, zp2, Pj.z);
+  fp_mul(xpzp, Pj.x, Pj.z);
+  fp2_mul_fp(xi_inv_zp3, xi_inv, zp3);
   fp12m *cur = &out, *tm = &tmp;
-  fp xp = Pa.x, yp = Pa.y;
+  fp xp = xpzp, yp = Pj.y; // a5 gets Xp*Zp, a0 gets Yp
   for (int i = 62; i >= 0; i--) {
     f12_sqr_nn(*tm, *cur);
     {
@@ -1249,7 +1275,7 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
       fp2_mul_small(t, t, 3);
       fp2_dbl(t2, Y2);
       fp2_sub(t, t, t2);
-      fp2_mul(a3, t, xi_inv);
+      fp2_mul(a3, t, xi_inv_zp3);
       fp2_mul(t, X2, Z2);
       fp2_mul_small(t, t, 3);
       fp2_mul_fp(t, t, xp);
@@ -1277,7 +1303,7 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1a &Pa,
       fp2_mul(t, M, T.x);
       fp2_mul(t2, T.y, H);
       fp2_sub(t, t, t2);
-      fp2_mul(a3, t, xi_inv);
+      fp2_mul(a3, t, xi_inv_zp3);
       fp2_mul(t, M, Z2);
       fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);

@@ -33,7 +33,7 @@ struct BlsWork {
   g1j *apk;        // [n] precomputed aggregate pubkeys (k>1 sets)
   uint64_t *agg_idx; // [n] indices of k>1 sets (count in agg_count[0])
   uint32_t *agg_count;
-  g1a *p_scaled;   // [n] r_i * aggregate pubkey (affine)
+  g1j *p_scaled;   // [n] r_i * aggregate pubkey (Jacobian)
   g2a *h2c;        // [n] hash_to_curve(msg)
   g2j *rsig;       // [n] r_i * sigma (jacobian)
   fp12m *fparts;   // [n] per-set miller values
@@ -167,11 +167,9 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
   uint8_t rbe[8];
 #pragma unroll
   for (int b = 0; b < 8; b++) rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
-  g1a apk_a;
-  g1j_to_aff(apk_a, apk);
   g1j rp;
-  g1j_mul_be(rp, apk_a, rbe, 8);
-  g1j_to_aff(w.p_scaled[i], rp);
+  g1j_mul_be_j(rp, apk, rbe, 8); // P stays Jacobian end-to-end
+  w.p_scaled[i] = rp;
   if (sig.inf) {
     // infinity is a valid subgroup element; contributes nothing
     fp2_zero(w.rsig[i].x);
@@ -364,7 +362,7 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t off_apk = bytes; bytes += align(n * sizeof(g1j));
   uint64_t off_aggidx = bytes; bytes += align(n * 8);
   uint64_t off_aggcnt = bytes; bytes += 256;
-  uint64_t off_p = bytes; bytes += align(n * sizeof(g1a));
+  uint64_t off_p = bytes; bytes += align(n * sizeof(g1j));
   uint64_t off_h = bytes; bytes += align(n * sizeof(g2a));
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_f = bytes; bytes += align(n * sizeof(fp12m));
@@ -381,7 +379,7 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.apk = reinterpret_cast<g1j *>(base + off_apk);
   w.agg_idx = reinterpret_cast<uint64_t *>(base + off_aggidx);
   w.agg_count = reinterpret_cast<uint32_t *>(base + off_aggcnt);
-  w.p_scaled = reinterpret_cast<g1a *>(base + off_p);
+  w.p_scaled = reinterpret_cast<g1j *>(base + off_p);
   w.h2c = reinterpret_cast<g2a *>(base + off_h);
   w.rsig = reinterpret_cast<g2j *>(base + off_r);
   w.fparts = reinterpret_cast<fp12m *>(base + off_f);
