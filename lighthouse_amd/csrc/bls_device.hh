@@ -1234,18 +1234,21 @@ __device__ inline bool f12_is_one(const fp12m &a) {
 // tmp must be distinct. Inversion-free Jacobian loop on the twist with
 // sparse lines (formulas validated against the Python reference).
 __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1j &Pj,
-                                  const g2a &Qa) {
+                                  const g2j &Qj) {
   f12_one(out);
-  if (fp_is_zero(Pj.z) || Qa.inf) return; // e(O,.) = e(.,O) = 1
-  g2j T;
-  g2j_from_aff(T, Qa);
-  // P stays JACOBIAN: lines scaled by the Fp-subfield factor Zp^3
-  // (final-exp invariant; validated in Python): a0 uses Yp directly,
-  // a3's xi^-1 absorbs Zp^3, a5 uses Xp*Zp — identical per-step cost and
-  // no P to-affine inversion. xi^-1 = (1-u)/2 is a constant.
-  fp2 xi_inv, xi_inv_zp3;
+  if (fp_is_zero(Pj.z) || fp2_is_zero(Qj.z)) return; // e(O,.) = e(.,O) = 1
+  g2j T = Qj;
+  // BOTH points stay JACOBIAN (no inversions anywhere; validated in
+  // Python). P side: lines scaled by the Fp-subfield factor Zp^3 — a0
+  // uses Yp directly, a3's xi^-1 absorbs Zp^3, a5 uses Xp*Zp. Q side:
+  // addition lines scaled by the further subfield-norm factor Zq^3 via
+  // H* = X*Zq^2 - Xq*Z^2 (= H*Zq^2), M* = Y*Zq^3 - Yq*Z^3 (= M*Zq^3),
+  // with a0/a3 consuming H*Zq. xi^-1 = (1-u)/2 is a constant.
+  fp2 xi_inv, xi_inv_zp3, zq2, zq3;
   FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
   fp_neg(xi_inv.c1, xi_inv.c0);
+  fp2_sqr(zq2, Qj.z);
+  fp2_mul(zq3, zq2, Qj.z);
   fp zp2, zp3, xpzp;
   fp_sqr(zp2, Pj.z);
   fp_mul(zp3, zp2, Pj.z);
@@ -1290,21 +1293,24 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1j &Pj,
       g2j_dbl(T, T);
     }
     if ((BLS_X_ABS >> i) & 1) {
-      // addition line through T (jac) and Q (affine)
-      fp2 Z2, Z3, H, M, a0, a3, a5, t, t2;
+      // addition line through T (jac) and Q (jac), scaled by Zq^3
+      fp2 Z2, Z3, Hs, Ms, HZq, a0, a3, a5, t, t2;
       fp2_sqr(Z2, T.z);
       fp2_mul(Z3, Z2, T.z);
-      fp2_mul(t, Qa.x, Z2);
-      fp2_sub(H, T.x, t);
-      fp2_mul(t, Qa.y, Z3);
-      fp2_sub(M, T.y, t);
-      fp2_mul(t, Z3, H);
+      fp2_mul(t, T.x, zq2);
+      fp2_mul(t2, Qj.x, Z2);
+      fp2_sub(Hs, t, t2); // H* = H*Zq^2
+      fp2_mul(t, T.y, zq3);
+      fp2_mul(t2, Qj.y, Z3);
+      fp2_sub(Ms, t, t2); // M* = M*Zq^3
+      fp2_mul(HZq, Hs, Qj.z);
+      fp2_mul(t, Z3, HZq);
       fp2_mul_fp(a0, t, yp);
-      fp2_mul(t, M, T.x);
-      fp2_mul(t2, T.y, H);
+      fp2_mul(t, Ms, T.x);
+      fp2_mul(t2, T.y, HZq);
       fp2_sub(t, t, t2);
       fp2_mul(a3, t, xi_inv_zp3);
-      fp2_mul(t, M, Z2);
+      fp2_mul(t, Ms, Z2);
       fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
@@ -1314,9 +1320,7 @@ __device__ inline void miller_raw(fp12m &out, fp12m &tmp, const g1j &Pj,
         cur = tm;
         tm = sw;
       }
-      g2j qj;
-      g2j_from_aff(qj, Qa);
-      g2j_add(T, T, qj);
+      g2j_add(T, T, Qj);
     }
   }
   f12_conj6_ip(*cur); // x < 0
@@ -1541,18 +1545,22 @@ struct miller_ws {
 // T-chain (lane 0) storing pre-step points, (2) line coefficients for all
 // steps computed in PARALLEL across lanes, (3) the f-chain with
 // cooperative fp12 ops consuming the stored coefficients.
-__device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
+__device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2j &Qj,
                                 f12w_ws &ws, miller_ws &mws, int lane) {
   if (lane == 0) f12_one(out);
   f12w_sync();
-  if (Pa.inf || Qa.inf) return;
+  if (Pa.inf || fp2_is_zero(Qj.z)) return;
   fp2 xi_inv; // constant (1-u)/2
   FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
   fp_neg(xi_inv.c1, xi_inv.c0);
+  // Q stays Jacobian: addition lines carry the extra subfield-norm factor
+  // Zq^3 (H* / M* / H*Zq form, as miller_raw; validated in Python)
+  fp2 zq2, zq3;
+  fp2_sqr(zq2, Qj.z);
+  fp2_mul(zq3, zq2, Qj.z);
   // ---- phase 1: serial point chain (lane 0 writes pre-step T) ----
   if (lane == 0) {
-    g2j T;
-    g2j_from_aff(T, Qa);
+    g2j T = Qj;
     int idx = 0;
     for (int i = 62; i >= 0; i--) {
       mws.T[idx][0] = T.x;
@@ -1565,9 +1573,7 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
         mws.T[idx][1] = T.y;
         mws.T[idx][2] = T.z;
         idx++;
-        g2j qj;
-        g2j_from_aff(qj, Qa);
-        g2j_add(T, T, qj);
+        g2j_add(T, T, Qj);
       }
     }
   }
@@ -1612,20 +1618,23 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2a &Qa,
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
     } else {
-      fp2 Z2, Z3, H, M;
+      fp2 Z2, Z3, Hs, Ms, HZq;
       fp2_sqr(Z2, T.z);
       fp2_mul(Z3, Z2, T.z);
-      fp2_mul(t, Qa.x, Z2);
-      fp2_sub(H, T.x, t);
-      fp2_mul(t, Qa.y, Z3);
-      fp2_sub(M, T.y, t);
-      fp2_mul(t, Z3, H);
+      fp2_mul(t, T.x, zq2);
+      fp2_mul(t2, Qj.x, Z2);
+      fp2_sub(Hs, t, t2); // H* = H*Zq^2
+      fp2_mul(t, T.y, zq3);
+      fp2_mul(t2, Qj.y, Z3);
+      fp2_sub(Ms, t, t2); // M* = M*Zq^3
+      fp2_mul(HZq, Hs, Qj.z);
+      fp2_mul(t, Z3, HZq);
       fp2_mul_fp(a0, t, Pa.y);
-      fp2_mul(t, M, T.x);
-      fp2_mul(t2, T.y, H);
+      fp2_mul(t, Ms, T.x);
+      fp2_mul(t2, T.y, HZq);
       fp2_sub(t, t, t2);
       fp2_mul(a3, t, xi_inv);
-      fp2_mul(t, M, Z2);
+      fp2_mul(t, Ms, Z2);
       fp2_mul_fp(t, t, Pa.x);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
@@ -1917,7 +1926,7 @@ __device__ inline void clear_cofactor_g2j(g2j &out, const g2j &p) {
   out = acc;
 }
 
-__device__ inline void h2c_g2(g2a &r, const uint8_t *msg) {
+__device__ inline void h2c_g2(g2j &r, const uint8_t *msg) {
   uint8_t uni[256];
   expand_message_xmd32(msg, uni);
   fp2 u0, u1;
@@ -1930,12 +1939,12 @@ __device__ inline void h2c_g2(g2a &r, const uint8_t *msg) {
   sswu_g2(q1p, u1);
   iso_map_g2(q0, q0p);
   iso_map_g2(q1, q1p);
-  g2j s, t, cleared;
+  g2j s, t;
   g2j_from_aff(s, q0);
   g2j_from_aff(t, q1);
   g2j_add(s, s, t);
-  clear_cofactor_g2j(cleared, s);
-  g2j_to_aff(r, cleared); // single inversion, at the very end
+  clear_cofactor_g2j(r, s); // stays Jacobian: consumers (Q-Jacobian
+                            // Miller loops) need no inversion at all
 }
 
 // G1 generator (negated y variant computed by callers when needed)

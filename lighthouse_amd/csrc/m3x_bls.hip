@@ -34,7 +34,7 @@ struct BlsWork {
   uint64_t *agg_idx; // [n] indices of k>1 sets (count in agg_count[0])
   uint32_t *agg_count;
   g1j *p_scaled;   // [n] r_i * aggregate pubkey (Jacobian)
-  g2a *h2c;        // [n] hash_to_curve(msg)
+  g2j *h2c;        // [n] hash_to_curve(msg), Jacobian (no inversion)
   g2j *rsig;       // [n] r_i * sigma (jacobian)
   fp12m *fparts;   // [n] per-set miller values
   int *fail;       // [1]
@@ -289,14 +289,13 @@ __global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
     if (lane == 0) *w.verdict = 0;
     return;
   }
-  g2a sig_a; // all lanes redundantly (wave-uniform)
-  g2j_to_aff(sig_a, w.sig_sum[0]);
   g1a ng1;
   g1_gen(ng1);
   fp_neg(ng1.y, ng1.y);
   if (lane == 0) f12_copy(sh[0], w.gt_parts[0]);
   f12w_sync();
-  miller_w(sh[1], ng1, sig_a, ws, mws, lane);
+  // sig_sum stays Jacobian: the Q-Jacobian Miller loop needs no inversion
+  miller_w(sh[1], ng1, w.sig_sum[0], ws, mws, lane);
   f12_mul_w(sh[1], sh[0], sh[1], ws, lane); // f_total
   final_exp_w(sh[2], sh[1], &sh[3], ws, lane);
   if (lane == 0) *w.verdict = f12_is_one(sh[2]) ? 1 : 0;
@@ -363,7 +362,7 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t off_aggidx = bytes; bytes += align(n * 8);
   uint64_t off_aggcnt = bytes; bytes += 256;
   uint64_t off_p = bytes; bytes += align(n * sizeof(g1j));
-  uint64_t off_h = bytes; bytes += align(n * sizeof(g2a));
+  uint64_t off_h = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_f = bytes; bytes += align(n * sizeof(fp12m));
   uint64_t off_fail = bytes; bytes += 256;
@@ -380,7 +379,7 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.agg_idx = reinterpret_cast<uint64_t *>(base + off_aggidx);
   w.agg_count = reinterpret_cast<uint32_t *>(base + off_aggcnt);
   w.p_scaled = reinterpret_cast<g1j *>(base + off_p);
-  w.h2c = reinterpret_cast<g2a *>(base + off_h);
+  w.h2c = reinterpret_cast<g2j *>(base + off_h);
   w.rsig = reinterpret_cast<g2j *>(base + off_r);
   w.fparts = reinterpret_cast<fp12m *>(base + off_f);
   w.fail = reinterpret_cast<int *>(base + off_fail);
