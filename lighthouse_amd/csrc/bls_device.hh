@@ -1374,20 +1374,15 @@ __device__ inline void f12_mul_w(fp12m &r, const fp12m &a, const fp12m &b,
 __device__ inline void f12_line_w(fp12m &f, const fp2 &a0, const fp2 &a3,
                                   const fp2 &a5, f12w_ws &ws, int lane) {
   if (lane < 18) {
+    // ONE uniform fp2_mul: per-branch multiplies would execute all three
+    // bodies serially in the wave (see f12_sqr_w note)
     int k = lane / 3, q = lane % 3;
-    fp2 fk, t;
-    if (q == 0) {
-      f12_get(f, k, fk);
-      fp2_mul(t, fk, a0);
-    } else if (q == 1) {
-      f12_get(f, (k + 3) % 6, fk);
-      fp2_mul(t, fk, a3);
-      if (k < 3) fp2_mul_xi(t, t);
-    } else {
-      f12_get(f, (k + 1) % 6, fk);
-      fp2_mul(t, fk, a5);
-      if (k < 5) fp2_mul_xi(t, t);
-    }
+    int src = q == 0 ? k : (q == 1 ? (k + 3) % 6 : (k + 1) % 6);
+    fp2 fk, coef, t;
+    f12_get(f, src, fk);
+    coef = q == 0 ? a0 : (q == 1 ? a3 : a5);
+    fp2_mul(t, fk, coef);
+    if ((q == 1 && k < 3) || (q == 2 && k < 5)) fp2_mul_xi(t, t);
     ws.t[lane] = t;
   }
   f12w_sync();
@@ -1488,14 +1483,13 @@ __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
   int i = F12_SQR_PI[li];
   int j = F12_SQR_PJ[li];
   if (lane < 21) {
+    // UNIFORM fp2_mul even on the diagonal: a divergent fp2_sqr branch
+    // makes the wave execute BOTH bodies (5 fp_mul issue slots instead of
+    // 3) — measured 35us vs 19us per op (tools/finishbench)
     fp2 ai, aj, t;
     f12_get(a, i, ai);
-    if (i == j) {
-      fp2_sqr(t, ai);
-    } else {
-      f12_get(a, j, aj);
-      fp2_mul(t, ai, aj);
-    }
+    f12_get(a, j, aj);
+    fp2_mul(t, ai, aj);
     ws.t[lane] = t;
   }
   f12w_sync();
@@ -1545,19 +1539,27 @@ struct miller_ws {
 // T-chain (lane 0) storing pre-step points, (2) line coefficients for all
 // steps computed in PARALLEL across lanes, (3) the f-chain with
 // cooperative fp12 ops consuming the stored coefficients.
-__device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2j &Qj,
+__device__ inline void miller_w(fp12m &out, const g1j &Pj, const g2j &Qj,
                                 f12w_ws &ws, miller_ws &mws, int lane) {
   if (lane == 0) f12_one(out);
   f12w_sync();
-  if (Pa.inf || fp2_is_zero(Qj.z)) return;
+  if (fp_is_zero(Pj.z) || fp2_is_zero(Qj.z)) return;
   fp2 xi_inv; // constant (1-u)/2
   FP_LOAD_C(xi_inv.c0, FP_TWO_INV);
   fp_neg(xi_inv.c1, xi_inv.c0);
-  // Q stays Jacobian: addition lines carry the extra subfield-norm factor
-  // Zq^3 (H* / M* / H*Zq form, as miller_raw; validated in Python)
-  fp2 zq2, zq3;
+  // BOTH points Jacobian (as miller_raw; validated in Python). P side:
+  // lines scaled by Zp^3 (a0 <- Yp, a3's xi^-1 absorbs Zp^3, a5 <- Xp*Zp).
+  // Q side: addition lines carry the further subfield-norm factor Zq^3
+  // (H* / M* / H*Zq form). Wave-uniform values, computed redundantly.
+  fp2 zq2, zq3, xi_inv_zp3;
   fp2_sqr(zq2, Qj.z);
   fp2_mul(zq3, zq2, Qj.z);
+  fp zp2, zp3, xp, yp;
+  fp_sqr(zp2, Pj.z);
+  fp_mul(zp3, zp2, Pj.z);
+  fp_mul(xp, Pj.x, Pj.z);
+  yp = Pj.y;
+  fp2_mul_fp(xi_inv_zp3, xi_inv, zp3);
   // ---- phase 1: serial point chain (lane 0 writes pre-step T) ----
   if (lane == 0) {
     g2j T = Qj;
@@ -1606,15 +1608,15 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2j &Qj,
       fp2_mul(Z3, Z2, T.z);
       fp2_mul(t, T.y, Z3);
       fp2_dbl(t, t);
-      fp2_mul_fp(a0, t, Pa.y);
+      fp2_mul_fp(a0, t, yp);
       fp2_mul(t, X2, T.x);
       fp2_mul_small(t, t, 3);
       fp2_dbl(t2, Y2);
       fp2_sub(t, t, t2);
-      fp2_mul(a3, t, xi_inv);
+      fp2_mul(a3, t, xi_inv_zp3);
       fp2_mul(t, X2, Z2);
       fp2_mul_small(t, t, 3);
-      fp2_mul_fp(t, t, Pa.x);
+      fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
     } else {
@@ -1629,13 +1631,13 @@ __device__ inline void miller_w(fp12m &out, const g1a &Pa, const g2j &Qj,
       fp2_sub(Ms, t, t2); // M* = M*Zq^3
       fp2_mul(HZq, Hs, Qj.z);
       fp2_mul(t, Z3, HZq);
-      fp2_mul_fp(a0, t, Pa.y);
+      fp2_mul_fp(a0, t, yp);
       fp2_mul(t, Ms, T.x);
       fp2_mul(t2, T.y, HZq);
       fp2_sub(t, t, t2);
-      fp2_mul(a3, t, xi_inv);
+      fp2_mul(a3, t, xi_inv_zp3);
       fp2_mul(t, Ms, Z2);
-      fp2_mul_fp(t, t, Pa.x);
+      fp2_mul_fp(t, t, xp);
       fp2_neg(t, t);
       fp2_mul(a5, t, xi_inv);
     }

@@ -210,6 +210,28 @@ __global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ m
   m3xb::h2c_g2(w.h2c[i], msgs + 32 * i);
 }
 
+// small-batch variant: ONE WAVE PER SET (cooperative miller_w). At tiny n
+// (block import: ~131 sets) the per-lane kernel is latency-bound — a
+// single set's serial Miller loop is tens of ms on one lane — while n
+// cooperative waves spread across 256 CUs cut that ~10x. Crossover is
+// empirical (M3X_SMALL_MILLER sets the threshold).
+__global__ __launch_bounds__(64) void k_bls_miller_small(uint64_t n,
+                                                         BlsWork w) {
+  __shared__ fp12m f;
+  __shared__ f12w_ws ws;
+  __shared__ miller_ws mws;
+  uint64_t i = blockIdx.x;
+  int lane = threadIdx.x;
+  if (i >= n) return;
+  if (*w.fail) {
+    if (lane == 0) f12_one(w.fparts[i]);
+    return;
+  }
+  miller_w(f, w.p_scaled[i], w.h2c[i], ws, mws, lane);
+  __syncthreads();
+  if (lane == 0) w.fparts[i] = f;
+}
+
 __global__ __launch_bounds__(64, 1) void k_bls_miller(uint64_t n, BlsWork w) {
   // fp12 state stays in thread-local scratch: an LDS-resident variant
   // measured 2x SLOWER (123ms vs 63ms on C2) — the L1/L2-cached spill
@@ -289,9 +311,14 @@ __global__ __launch_bounds__(64) void k_bls_finish(BlsWork w) {
     if (lane == 0) *w.verdict = 0;
     return;
   }
-  g1a ng1;
-  g1_gen(ng1);
-  fp_neg(ng1.y, ng1.y);
+  g1j ng1;
+  {
+    g1a g;
+    g1_gen(g);
+    ng1.x = g.x;
+    fp_neg(ng1.y, g.y);
+    fp_one(ng1.z);
+  }
   if (lane == 0) f12_copy(sh[0], w.gt_parts[0]);
   f12w_sync();
   // sig_sum stays Jacobian: the Q-Jacobian Miller loop needs no inversion
@@ -413,8 +440,15 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream, ctx->ev_s2, 0));
   DBG_STEP(ctx, "h2c");
   m3x::time_begin(ctx, M3X_K_BLS_MILLER);
-  hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream, n,
-                     w);
+  // small batches are latency-bound on the per-lane kernel: go wave-per-set
+  uint64_t small_thresh = 2048; // measured crossover (tmp_bench/c4probe)
+  if (const char *e = getenv("M3X_SMALL_MILLER")) small_thresh = strtoull(e, nullptr, 10);
+  if (n <= small_thresh)
+    hipLaunchKernelGGL(k_bls_miller_small, dim3((uint32_t)n), dim3(64), 0,
+                       ctx->stream, n, w);
+  else
+    hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream,
+                       n, w);
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
   DBG_STEP(ctx, "miller");
   uint32_t rblocks = (uint32_t)((n + 255) / 256);

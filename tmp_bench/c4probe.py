@@ -70,6 +70,15 @@ for lbl, thr in (("per-lane", "0"), ("wave-per-set", "100000")):
     os.environ["M3X_SMALL_MILLER"] = thr
     print(f"C4 131 sets  {lbl:13s}: {run(d4):8.2f} ms", flush=True)
 
+# per-kernel split of the wave-per-set C4 run (HIP-event timing; the
+# timed runs above keep timing OFF to preserve stream overlap)
+os.environ["M3X_SMALL_MILLER"] = "100000"
+ctx.timing_enable(True)
+run(d4, reps=3)
+print("C4 kernel split:", {k: round(v, 2) for k, v in ctx.kernel_times().items()},
+      flush=True)
+ctx.timing_enable(False)
+
 # threshold sweep on k=1 sets
 for n in (512, 2048, 8192, 16384):
     idx = list(range(n))

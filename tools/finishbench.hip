@@ -74,6 +74,47 @@ __global__ __launch_bounds__(64) void k_finishbench(uint64_t *stamp,
   STAMP(5); // 64 cooperative generic muls
   for (int i = 0; i < 64; i++) f12_sqr_w(sh[5], ws, lane);
   STAMP(6); // 64 cooperative squarings
+  // product phase only (lane<36 fp2_mul into ws) — isolates the math
+  for (int i = 0; i < 64; i++) {
+    if (lane < 36) {
+      int pi = lane / 6, pj = lane % 6;
+      fp2 ai, bj, t;
+      f12_get(sh[5], pi, ai);
+      f12_get(sh[1], pj, bj);
+      fp2_mul(t, ai, bj);
+      ws.t[lane] = t;
+    }
+    __syncthreads();
+  }
+  STAMP(7);
+  // accumulate phase only (lane<6 fold of ws.t)
+  for (int it = 0; it < 64; it++) {
+    if (lane < 6) {
+      fp2 acc, h, t;
+      fp2_zero(acc);
+      fp2_zero(h);
+      for (int i = 0; i <= lane; i++) {
+        t = ws.t[i * 6 + (lane - i)];
+        fp2_add(acc, acc, t);
+      }
+      for (int i = lane + 1; i < 6; i++) {
+        t = ws.t[i * 6 + (lane + 6 - i)];
+        fp2_add(h, h, t);
+      }
+      fp2_mul_xi(h, h);
+      fp2_add(acc, acc, h);
+      f12_set(sh[5], lane, acc);
+    }
+    __syncthreads();
+  }
+  STAMP(8);
+  // bare fp2_mul, all 64 lanes, no LDS (pure issue cost reference)
+  {
+    fp2 x = Qj.x, y = Qj.y;
+    for (int i = 0; i < 64; i++) fp2_mul(x, x, y);
+    if (lane == 0) sh[6].s[2] = x.c0;
+  }
+  STAMP(9);
   // phase-1-equivalent: the serial T-chain alone on lane 0
   if (lane == 0) {
     g2j T = Qj;
@@ -94,14 +135,14 @@ __global__ __launch_bounds__(64) void k_finishbench(uint64_t *stamp,
     }
     sh[6].s[0] = T.x.c0; // keep alive
   }
-  STAMP(7);
+  STAMP(10);
   // serial fp_inv on lane 0 (the f12_inv_w norm tail)
   if (lane == 0) {
     fp v = sh[0].s[1], r;
     fp_inv(r, v);
     sh[6].s[1] = r;
   }
-  STAMP(8);
+  STAMP(11);
   if (lane == 0) {
     for (int i = 0; i < 7; i++) f12_copy(out[i], sh[i]);
   }
@@ -120,17 +161,19 @@ int main() {
   (void)hipDeviceSynchronize();
   uint64_t st[NSTAMP];
   (void)hipMemcpy(st, stamp_d, NSTAMP * 8, hipMemcpyDeviceToHost);
-  const char *names[] = {"miller_w (full)",  "final_exp_w (full)",
-                         "f12_inv_w",        "f12_pow_xabs_w",
-                         "64x f12_mul_w",    "64x f12_sqr_w",
-                         "serial T-chain",   "serial fp_inv"};
+  const char *names[] = {"miller_w (full)",   "final_exp_w (full)",
+                         "f12_inv_w",         "f12_pow_xabs_w",
+                         "64x f12_mul_w",     "64x f12_sqr_w",
+                         "64x product-only",  "64x accumulate-only",
+                         "64x bare fp2_mul",  "serial T-chain",
+                         "serial fp_inv"};
   // wall_clock64 is the constant ~100 MHz counter
   double mhz = 100.0;
-  for (int k = 0; k < 8; k++) {
+  for (int k = 0; k < 11; k++) {
     double ms = (double)(st[k + 1] - st[k]) / (mhz * 1000.0);
     printf("%-20s %10.3f ms\n", names[k], ms);
   }
   printf("total                %10.3f ms\n",
-         (double)(st[8] - st[0]) / (mhz * 1000.0));
+         (double)(st[11] - st[0]) / (mhz * 1000.0));
   return 0;
 }
