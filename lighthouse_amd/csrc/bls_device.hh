@@ -1344,27 +1344,33 @@ __device__ inline void f12_mul_w(fp12m &r, const fp12m &a, const fp12m &b,
                                  f12w_ws &ws, int lane) {
   if (lane < 36) {
     int i = lane / 6, j = lane % 6;
+    // xi-weighting happens HERE (lane-parallel) and products land in
+    // residue-grouped slots (class k=(i+j)%6 has exactly 6 entries at
+    // k*6+i), so the fold is a short multi-lane tree: the old 6-lane
+    // serial accumulate cost MORE than the whole 36-lane product phase
+    // (12us vs 8.7us, tools/finishbench).
     fp2 ai, bj, t;
     f12_get(a, i, ai);
     f12_get(b, j, bj);
     fp2_mul(t, ai, bj);
-    ws.t[lane] = t;
+    if (i + j >= 6) fp2_mul_xi(t, t);
+    ws.t[((i + j) % 6) * 6 + i] = t;
+  }
+  f12w_sync();
+  if (lane < 18) { // fold 6 -> 3 per class
+    int k = lane / 3, pos = lane % 3;
+    fp2 t0 = ws.t[k * 6 + pos], t1 = ws.t[k * 6 + pos + 3];
+    fp2_add(t0, t0, t1);
+    ws.t[k * 6 + pos] = t0;
   }
   f12w_sync();
   if (lane < 6) {
-    fp2 acc, h, t;
-    fp2_zero(acc);
-    fp2_zero(h);
-    for (int i = 0; i <= lane; i++) {
-      t = ws.t[i * 6 + (lane - i)];
-      fp2_add(acc, acc, t);
-    }
-    for (int i = lane + 1; i < 6; i++) {
-      t = ws.t[i * 6 + (lane + 6 - i)];
-      fp2_add(h, h, t);
-    }
-    fp2_mul_xi(h, h);
-    fp2_add(acc, acc, h);
+    fp2 acc, t;
+    acc = ws.t[lane * 6];
+    t = ws.t[lane * 6 + 1];
+    fp2_add(acc, acc, t);
+    t = ws.t[lane * 6 + 2];
+    fp2_add(acc, acc, t);
     f12_set(r, lane, acc);
   }
   f12w_sync();
@@ -1473,6 +1479,14 @@ __device__ inline void f12_inv_w(fp12m &r, const fp12m &a, fp12m &g, fp12m &t,
 }
 
 // cooperative squaring: 21 distinct products across lanes (alias-safe)
+// residue-grouped slot map for the cooperative squaring fold: product
+// qi (pi<=pj enumeration) lands at slot ((pi+pj)%6)*4 + pos; groups have
+// 3 or 4 entries, pads (slots 7, 15, 23) are zeroed by lanes 21-23.
+// Doubling (pi!=pj) and xi-weighting (pi+pj>=6) are applied in the
+// product phase.
+__constant__ int F12_SQR_SLOT[21] = {0, 4,  8,  12, 16, 20, 9, 13, 17,
+                                     21, 1, 18, 22, 2,  5,  3, 6,  10,
+                                     11, 14, 19};
 __constant__ int F12_SQR_PI[21] = {0, 0, 0, 0, 0, 0, 1, 1, 1, 1, 1,
                                    2, 2, 2, 2, 3, 3, 3, 4, 4, 5};
 __constant__ int F12_SQR_PJ[21] = {0, 1, 2, 3, 4, 5, 1, 2, 3, 4, 5,
@@ -1483,40 +1497,40 @@ __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
   int i = F12_SQR_PI[li];
   int j = F12_SQR_PJ[li];
   if (lane < 21) {
-    // UNIFORM fp2_mul even on the diagonal: a divergent fp2_sqr branch
-    // makes the wave execute BOTH bodies (5 fp_mul issue slots instead of
-    // 3) — measured 35us vs 19us per op (tools/finishbench)
+    // UNIFORM fp2_mul even on the diagonal (a divergent fp2_sqr branch
+    // would execute both bodies); doubling + xi-weighting lane-parallel
+    // here; residue-grouped slots so the fold is a 2-step tree — the old
+    // 21-entry 6-lane scan was ~27us, 3x the product phase.
     fp2 ai, aj, t;
     f12_get(a, i, ai);
     f12_get(a, j, aj);
     fp2_mul(t, ai, aj);
-    ws.t[lane] = t;
+    if (i != j) fp2_dbl(t, t);
+    if (i + j >= 6) fp2_mul_xi(t, t);
+    ws.t[F12_SQR_SLOT[lane]] = t;
+  } else if (lane < 24) { // zero the pad slots 7, 15, 23
+    fp2 z;
+    fp2_zero(z);
+    ws.t[8 * (lane - 21) + 7] = z;
+  }
+  f12w_sync();
+  if (lane < 12) { // fold 4 -> 2 per class
+    int k = lane / 2, pos = lane % 2;
+    fp2 t0 = ws.t[k * 4 + pos], t1 = ws.t[k * 4 + pos + 2];
+    fp2_add(t0, t0, t1);
+    ws.t[k * 4 + pos] = t0;
   }
   f12w_sync();
   if (lane < 6) {
-    fp2 acc, hacc, t;
-    fp2_zero(acc);
-    fp2_zero(hacc);
-    int qi = 0;
-    for (int pi = 0; pi < 6; pi++) {
-      for (int pj = pi; pj < 6; pj++, qi++) {
-        int sum = pi + pj;
-        if (sum == lane || sum == lane + 6) {
-          t = ws.t[qi];
-          if (pi != pj) fp2_dbl(t, t);
-          if (sum == lane)
-            fp2_add(acc, acc, t);
-          else
-            fp2_add(hacc, hacc, t);
-        }
-      }
-    }
-    fp2_mul_xi(hacc, hacc);
-    fp2_add(acc, acc, hacc);
+    fp2 acc, t;
+    acc = ws.t[lane * 4];
+    t = ws.t[lane * 4 + 1];
+    fp2_add(acc, acc, t);
     f12_set(a, lane, acc);
   }
   f12w_sync();
 }
+
 
 // r = a^|x|; r distinct from a
 __device__ inline void f12_pow_xabs_w(fp12m &r, const fp12m &a, f12w_ws &ws,
