@@ -268,6 +268,25 @@ __device__ inline bool fp_sqrt(fp &r, const fp &a) {
   return true;
 }
 
+// sqrt AND inverse-sqrt in ONE pow (p = 3 mod 4): u = a^((p-3)/4) is
+// 1/sqrt(a) for square a (u*s = a^((p-1)/2) = 1), s = u*a = a^((p+1)/4).
+// Lets callers replace a sqrt-then-invert pair (two 381-bit pows) with
+// one pow + one mul. Returns false for non-squares (s, u then invalid).
+__device__ inline bool fp_sqrt_ui(fp &s, fp &u, const fp &a) {
+  // (p-3)/4: p[0] = ...aaab so p-3 borrows nothing
+  uint64_t t[6] = {BLS_P[0] - 3, BLS_P[1], BLS_P[2],
+                   BLS_P[3], BLS_P[4], BLS_P[5]};
+  uint64_t e[6];
+#pragma unroll
+  for (int i = 0; i < 6; i++)
+    e[i] = (t[i] >> 2) | (i < 5 ? (t[i + 1] << 62) : 0);
+  fp_pow_limbs(u, a, e, 6);
+  fp_mul(s, u, a);
+  fp s2;
+  fp_sqr(s2, s);
+  return fp_eq(s2, a);
+}
+
 // standard-form compare against (p-1)/2 ("lexicographically largest")
 __device__ inline bool fp_gt_half(const fp &a) {
   uint64_t s[6];
@@ -427,7 +446,7 @@ __device__ inline bool fp2_sqrt(fp2 &r, const fp2 &a) {
     r.c1 = s;
     return true;
   }
-  fp n, s, d, x0, x1, t, inv2;
+  fp n, s, d, x0, x1, t, u, inv2;
   fp_sqr(n, a.c0);
   fp_sqr(t, a.c1);
   fp_add(n, n, t);
@@ -435,14 +454,15 @@ __device__ inline bool fp2_sqrt(fp2 &r, const fp2 &a) {
   FP_LOAD_C(inv2, FP_TWO_INV); // constant 2^-1 (generator-emitted)
   fp_add(d, a.c0, s);
   fp_mul(d, d, inv2);
-  if (!fp_sqrt(x0, d)) {
+  if (!fp_sqrt_ui(x0, u, d)) {
     fp_sub(d, a.c0, s);
     fp_mul(d, d, inv2);
-    if (!fp_sqrt(x0, d)) return false;
+    if (!fp_sqrt_ui(x0, u, d)) return false;
   }
-  fp_add(t, x0, x0);
-  fp_inv(t, t);
-  fp_mul(x1, a.c1, t);
+  // x1 = c1/(2*x0) = c1 * u * 2^-1 (u = 1/x0 from the fused pow — no
+  // separate fp_inv)
+  fp_mul(x1, a.c1, u);
+  fp_mul(x1, x1, inv2);
   fp2 cand, sq;
   cand.c0 = x0;
   cand.c1 = x1;

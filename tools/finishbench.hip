@@ -148,6 +148,47 @@ __global__ __launch_bounds__(64) void k_finishbench(uint64_t *stamp,
   }
 }
 
+// per-lane h2c latency breakdown (all 64 lanes compute the same message,
+// matching the per-lane production kernel's regime)
+__global__ __launch_bounds__(64) void k_h2cbench(uint64_t *stamp,
+                                                 g2j *out) {
+  int lane = threadIdx.x;
+  uint8_t msg[32];
+  for (int i = 0; i < 32; i++) msg[i] = (uint8_t)(i * 7 + 3);
+#define HSTAMP(k)                                                            \
+  do {                                                                       \
+    __syncthreads();                                                         \
+    if (lane == 0) stamp[k] = wall_clock64();                                \
+    __syncthreads();                                                         \
+  } while (0)
+  HSTAMP(0);
+  uint8_t uni[256];
+  expand_message_xmd32(msg, uni);
+  HSTAMP(1);
+  fp2 u0, u1;
+  h2f_from_be64(u0.c0, uni);
+  h2f_from_be64(u0.c1, uni + 64);
+  h2f_from_be64(u1.c0, uni + 128);
+  h2f_from_be64(u1.c1, uni + 192);
+  HSTAMP(2);
+  g2a q0p, q1p, q0, q1;
+  sswu_g2(q0p, u0);
+  sswu_g2(q1p, u1);
+  HSTAMP(3);
+  iso_map_g2(q0, q0p);
+  iso_map_g2(q1, q1p);
+  HSTAMP(4);
+  g2j s, t;
+  g2j_from_aff(s, q0);
+  g2j_from_aff(t, q1);
+  g2j_add(s, s, t);
+  HSTAMP(5);
+  g2j r;
+  clear_cofactor_g2j(r, s);
+  HSTAMP(6);
+  if (lane == 0) out[0] = r;
+}
+
 int main() {
   uint64_t *stamp_d;
   fp12m *out_d;
@@ -175,5 +216,20 @@ int main() {
   }
   printf("total                %10.3f ms\n",
          (double)(st[11] - st[0]) / (mhz * 1000.0));
+  // h2c per-lane latency breakdown
+  g2j *h2c_d;
+  (void)hipMalloc(&h2c_d, sizeof(g2j));
+  for (int it = 0; it < 3; it++)
+    hipLaunchKernelGGL(k_h2cbench, dim3(1), dim3(64), 0, 0, stamp_d, h2c_d);
+  (void)hipDeviceSynchronize();
+  (void)hipMemcpy(st, stamp_d, NSTAMP * 8, hipMemcpyDeviceToHost);
+  const char *hn[] = {"expand_message", "h2f x4", "sswu x2", "iso_map x2",
+                      "from_aff+add",   "clear_cofactor"};
+  printf("-- h2c per-lane latency --\n");
+  for (int k = 0; k < 6; k++)
+    printf("%-20s %10.3f ms\n", hn[k],
+           (double)(st[k + 1] - st[k]) / (mhz * 1000.0));
+  printf("h2c total            %10.3f ms\n",
+         (double)(st[6] - st[0]) / (mhz * 1000.0));
   return 0;
 }
