@@ -1807,25 +1807,20 @@ __device__ inline void expand_message_xmd32(const uint8_t *msg,
   }
 }
 
-__device__ inline void sswu_g2(g2a &out, const fp2 &u) {
-  fp2 A, B, Z;
+// SSWU tail given 1 + 1/tv already in hand (or the tv=0 constant case
+// signaled by tv_zero): shared by the single and dual entry points.
+__device__ inline void sswu_g2_tail(g2a &out, const fp2 &u, const fp2 &zu2,
+                                    const fp2 &inv_tv_p1, bool tv_zero) {
+  fp2 A, B;
   FP2_LOAD_C(A, SSWU_A);
   FP2_LOAD_C(B, SSWU_B);
-  FP2_LOAD_C(Z, SSWU_Z);
-  fp2 zu2, tv, x1, gx1, y1, x, y, t;
-  fp2_sqr(zu2, u);
-  fp2_mul(zu2, zu2, Z);
-  fp2_sqr(tv, zu2);
-  fp2_add(tv, tv, zu2);
-  if (fp2_is_zero(tv)) {
+  fp2 x1, gx1, y1, x, y, t;
+  if (tv_zero) {
     FP2_LOAD_C(x1, SSWU_B_DIV_ZA); // constant B/(Z*A)
   } else {
-    fp2 inv_tv, one, nb_over_a;
-    fp2_inv(inv_tv, tv);
-    fp2_one(one);
-    fp2_add(inv_tv, one, inv_tv);
+    fp2 nb_over_a;
     FP2_LOAD_C(nb_over_a, SSWU_NB_DIV_A); // constant -B/A
-    fp2_mul(x1, nb_over_a, inv_tv);
+    fp2_mul(x1, nb_over_a, inv_tv_p1);
   }
   fp2_sqr(gx1, x1);
   fp2_mul(gx1, gx1, x1);
@@ -1851,6 +1846,57 @@ __device__ inline void sswu_g2(g2a &out, const fp2 &u) {
   out.x = x;
   out.y = y;
   out.inf = 0;
+}
+
+__device__ inline void sswu_g2(g2a &out, const fp2 &u) {
+  fp2 Z, zu2, tv, inv_tv;
+  FP2_LOAD_C(Z, SSWU_Z);
+  fp2_sqr(zu2, u);
+  fp2_mul(zu2, zu2, Z);
+  fp2_sqr(tv, zu2);
+  fp2_add(tv, tv, zu2);
+  bool tz = fp2_is_zero(tv);
+  fp2_zero(inv_tv);
+  if (!tz) {
+    fp2 one;
+    fp2_inv(inv_tv, tv);
+    fp2_one(one);
+    fp2_add(inv_tv, one, inv_tv);
+  }
+  sswu_g2_tail(out, u, zu2, inv_tv, tz);
+}
+
+// both h2c points with ONE shared tv inversion (Montgomery batch): the
+// per-point fp2_inv is a full 381-bit pow, the batching costs 3 muls
+__device__ inline void sswu_g2_dual(g2a out[2], const fp2 u[2]) {
+  fp2 Z, zu2[2], tv[2];
+  FP2_LOAD_C(Z, SSWU_Z);
+  bool tz[2];
+  for (int i = 0; i < 2; i++) {
+    fp2_sqr(zu2[i], u[i]);
+    fp2_mul(zu2[i], zu2[i], Z);
+    fp2_sqr(tv[i], zu2[i]);
+    fp2_add(tv[i], tv[i], zu2[i]);
+    tz[i] = fp2_is_zero(tv[i]);
+  }
+  fp2 inv[2], one;
+  fp2_one(one);
+  if (!tz[0] && !tz[1]) {
+    fp2 prod, pinv;
+    fp2_mul(prod, tv[0], tv[1]);
+    fp2_inv(pinv, prod);
+    fp2_mul(inv[0], pinv, tv[1]);
+    fp2_mul(inv[1], pinv, tv[0]);
+  } else { // rare (crafted u): fall back per point
+    for (int i = 0; i < 2; i++) {
+      fp2_zero(inv[i]);
+      if (!tz[i]) fp2_inv(inv[i], tv[i]);
+    }
+  }
+  for (int i = 0; i < 2; i++) {
+    fp2_add(inv[i], inv[i], one);
+    sswu_g2_tail(out[i], u[i], zu2[i], inv[i], tz[i]);
+  }
 }
 
 __device__ inline void iso_map_g2(g2a &out, const g2a &in) {
@@ -1902,6 +1948,58 @@ __device__ inline void iso_map_g2(g2a &out, const g2a &in) {
   fp2_mul(out.y, yn, ydi);
   fp2_mul(out.y, out.y, in.y);
   out.inf = 0;
+}
+
+// iso_map emitting JACOBIAN coordinates — no inversion at all:
+// x = xn/xd, y = in.y*yn/yd maps to Z = xd*yd, X = xn*xd*yd^2,
+// Y = in.y*yn*yd^2*xd^3 (X/Z^2 = xn/xd, Y/Z^3 = in.y*yn/yd).
+__device__ inline void iso_map_g2_j(g2j &out, const g2a &in) {
+  fp2 k[4], xn, xd, yn, yd;
+  FP2_LOAD_C(k[0], ISO_XNUM0);
+  FP2_LOAD_C(k[1], ISO_XNUM1);
+  FP2_LOAD_C(k[2], ISO_XNUM2);
+  FP2_LOAD_C(k[3], ISO_XNUM3);
+  xn = k[3];
+  for (int i = 2; i >= 0; i--) {
+    fp2_mul(xn, xn, in.x);
+    fp2_add(xn, xn, k[i]);
+  }
+  FP2_LOAD_C(k[0], ISO_XDEN0);
+  FP2_LOAD_C(k[1], ISO_XDEN1);
+  FP2_LOAD_C(k[2], ISO_XDEN2);
+  xd = k[2];
+  for (int i = 1; i >= 0; i--) {
+    fp2_mul(xd, xd, in.x);
+    fp2_add(xd, xd, k[i]);
+  }
+  FP2_LOAD_C(k[0], ISO_YNUM0);
+  FP2_LOAD_C(k[1], ISO_YNUM1);
+  FP2_LOAD_C(k[2], ISO_YNUM2);
+  FP2_LOAD_C(k[3], ISO_YNUM3);
+  yn = k[3];
+  for (int i = 2; i >= 0; i--) {
+    fp2_mul(yn, yn, in.x);
+    fp2_add(yn, yn, k[i]);
+  }
+  FP2_LOAD_C(k[0], ISO_YDEN0);
+  FP2_LOAD_C(k[1], ISO_YDEN1);
+  FP2_LOAD_C(k[2], ISO_YDEN2);
+  FP2_LOAD_C(k[3], ISO_YDEN3);
+  yd = k[3];
+  for (int i = 2; i >= 0; i--) {
+    fp2_mul(yd, yd, in.x);
+    fp2_add(yd, yd, k[i]);
+  }
+  fp2 yd2, xd2, xd3, t;
+  fp2_sqr(yd2, yd);
+  fp2_sqr(xd2, xd);
+  fp2_mul(xd3, xd2, xd);
+  fp2_mul(out.z, xd, yd);
+  fp2_mul(t, xn, xd);
+  fp2_mul(out.x, t, yd2);
+  fp2_mul(t, yd2, xd3);
+  fp2_mul(t, t, yn);
+  fp2_mul(out.y, t, in.y);
 }
 
 __device__ inline void g2j_neg(g2j &r, const g2j &p) {
@@ -1965,19 +2063,16 @@ __device__ inline void clear_cofactor_g2j(g2j &out, const g2j &p) {
 __device__ inline void h2c_g2(g2j &r, const uint8_t *msg) {
   uint8_t uni[256];
   expand_message_xmd32(msg, uni);
-  fp2 u0, u1;
-  h2f_from_be64(u0.c0, uni);
-  h2f_from_be64(u0.c1, uni + 64);
-  h2f_from_be64(u1.c0, uni + 128);
-  h2f_from_be64(u1.c1, uni + 192);
-  g2a q0p, q1p, q0, q1;
-  sswu_g2(q0p, u0);
-  sswu_g2(q1p, u1);
-  iso_map_g2(q0, q0p);
-  iso_map_g2(q1, q1p);
+  fp2 u[2];
+  h2f_from_be64(u[0].c0, uni);
+  h2f_from_be64(u[0].c1, uni + 64);
+  h2f_from_be64(u[1].c0, uni + 128);
+  h2f_from_be64(u[1].c1, uni + 192);
+  g2a qp[2];
+  sswu_g2_dual(qp, u); // one shared tv inversion for both points
   g2j s, t;
-  g2j_from_aff(s, q0);
-  g2j_from_aff(t, q1);
+  iso_map_g2_j(s, qp[0]); // Jacobian iso: no inversion at all
+  iso_map_g2_j(t, qp[1]);
   g2j_add(s, s, t);
   clear_cofactor_g2j(r, s); // stays Jacobian: consumers (Q-Jacobian
                             // Miller loops) need no inversion at all

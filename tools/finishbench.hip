@@ -171,16 +171,14 @@ __global__ __launch_bounds__(64) void k_h2cbench(uint64_t *stamp,
   h2f_from_be64(u1.c0, uni + 128);
   h2f_from_be64(u1.c1, uni + 192);
   HSTAMP(2);
-  g2a q0p, q1p, q0, q1;
-  sswu_g2(q0p, u0);
-  sswu_g2(q1p, u1);
+  fp2 uu[2] = {u0, u1};
+  g2a qp[2];
+  sswu_g2_dual(qp, uu);
   HSTAMP(3);
-  iso_map_g2(q0, q0p);
-  iso_map_g2(q1, q1p);
-  HSTAMP(4);
   g2j s, t;
-  g2j_from_aff(s, q0);
-  g2j_from_aff(t, q1);
+  iso_map_g2_j(s, qp[0]);
+  iso_map_g2_j(t, qp[1]);
+  HSTAMP(4);
   g2j_add(s, s, t);
   HSTAMP(5);
   g2j r;
