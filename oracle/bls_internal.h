@@ -63,7 +63,11 @@ void fp12_one(fp12_t *r);
 void fp12_mul_(fp12_t *r, const fp12_t *a, const fp12_t *b);
 int fp12_is_one(const fp12_t *a);
 void miller(fp12_t *f, const g1_aff_t *p, const g2_aff_t *q); /* f *= ML(p,q) */
+/* slow reference forms (cross-check tests only) */
+void miller_affine_ref(fp12_t *f, const g1_aff_t *p, const g2_aff_t *q);
+int g2_in_subgroup_ref(const g2_aff_t *p);
 void final_exp(fp12_t *r, const fp12_t *f);
+void final_exp3(fp12_t *r, const fp12_t *f); /* cubed hard part; ==1-equivalent */
 void fp12_to_bytes(const fp12_t *a, uint8_t out[576]);
 
 /* hash-to-curve (RFC 9380, G2 suite, DST = blst.rs:15) */
