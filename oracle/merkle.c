@@ -111,7 +111,15 @@ void m3x_oracle_validator_leaf(const uint8_t ssz[121], uint8_t out[32]) {
   memcpy(c[5], ssz + 97, 8);
   memcpy(c[6], ssz + 105, 8);
   memcpy(c[7], ssz + 113, 8);
-  m3x_oracle_merkleize((const uint8_t *)c, 8, 3, out);
+  /* fixed 8-leaf tree, unrolled (no per-validator malloc — hot path) */
+  uint8_t l1[4][32], l2[2][32];
+  m3x_oracle_hash64(c[0], c[1], l1[0]);
+  m3x_oracle_hash64(c[2], c[3], l1[1]);
+  m3x_oracle_hash64(c[4], c[5], l1[2]);
+  m3x_oracle_hash64(c[6], c[7], l1[3]);
+  m3x_oracle_hash64(l1[0], l1[1], l2[0]);
+  m3x_oracle_hash64(l1[2], l1[3], l2[1]);
+  m3x_oracle_hash64(l2[0], l2[1], out);
 }
 
 void m3x_oracle_validator_registry_root(const uint8_t *ssz, uint64_t n,

@@ -51,13 +51,16 @@ static void compress(uint32_t st[8], const uint8_t block[64]) {
   st[4] += e; st[5] += f; st[6] += g; st[7] += h;
 }
 
+static void compress_blocks(uint32_t st[8], const uint8_t *data, size_t nblk);
+
 void m3x_oracle_sha256(const uint8_t *data, size_t len, uint8_t out[32]) {
   uint32_t st[8];
   memcpy(st, IV, sizeof(IV));
   size_t off = 0;
-  while (len - off >= 64) {
-    compress(st, data + off);
-    off += 64;
+  size_t nfull = len / 64;
+  if (nfull) {
+    compress_blocks(st, data, nfull);
+    off = nfull * 64;
   }
   uint8_t tail[128];
   size_t rem = len - off;
@@ -68,9 +71,7 @@ void m3x_oracle_sha256(const uint8_t *data, size_t len, uint8_t out[32]) {
   uint64_t bits = (uint64_t)len * 8;
   for (int i = 0; i < 8; i++)
     tail[tlen - 1 - i] = (uint8_t)(bits >> (8 * i));
-  compress(st, tail);
-  if (tlen == 128)
-    compress(st, tail + 64);
+  compress_blocks(st, tail, tlen / 64);
   for (int i = 0; i < 8; i++) {
     out[4 * i] = (uint8_t)(st[i] >> 24);
     out[4 * i + 1] = (uint8_t)(st[i] >> 16);
@@ -81,8 +82,113 @@ void m3x_oracle_sha256(const uint8_t *data, size_t len, uint8_t out[32]) {
 
 void m3x_oracle_hash64(const uint8_t left[32], const uint8_t right[32],
                        uint8_t out[32]) {
-  uint8_t buf[64];
+  /* two-to-one: block 1 = left||right, block 2 = the constant padding
+     block (0x80, zeros, bit length 512) */
+  static const uint8_t PAD512[64] = {
+      0x80, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+      0,    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+      0,    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+      0,    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0x02, 0x00};
+  uint32_t st[8];
+  memcpy(st, IV, sizeof(IV));
+  uint8_t buf[128];
   memcpy(buf, left, 32);
   memcpy(buf + 32, right, 32);
-  m3x_oracle_sha256(buf, 64, out);
+  memcpy(buf + 64, PAD512, 64);
+  compress_blocks(st, buf, 2);
+  for (int i = 0; i < 8; i++) {
+    out[4 * i] = (uint8_t)(st[i] >> 24);
+    out[4 * i + 1] = (uint8_t)(st[i] >> 16);
+    out[4 * i + 2] = (uint8_t)(st[i] >> 8);
+    out[4 * i + 3] = (uint8_t)st[i];
+  }
+}
+
+/* ---------------- SHA-NI (x86 SHA extensions) fast path ----------------
+ * Runtime-dispatched: engaged when the CPU reports SHA extensions (this
+ * survey container's Xeon and EPYC GPU-box hosts both have sha_ni).
+ * Same FIPS 180-4 function — pinned by the same NIST KATs + hashlib
+ * golden vectors as the portable path (tests run both). */
+#if defined(__x86_64__)
+#include <immintrin.h>
+
+__attribute__((target("sha,sse4.1,ssse3")))
+static void compress_ni(uint32_t st[8], const uint8_t *data, size_t nblk) {
+  const __m128i MASK =
+      _mm_set_epi64x(0x0c0d0e0f08090a0bULL, 0x0405060700010203ULL);
+  __m128i T0 = _mm_loadu_si128((const __m128i *)st);
+  __m128i T1 = _mm_loadu_si128((const __m128i *)(st + 4));
+  T0 = _mm_shuffle_epi32(T0, 0xB1);         /* CDAB */
+  T1 = _mm_shuffle_epi32(T1, 0x1B);         /* EFGH */
+  __m128i S0 = _mm_alignr_epi8(T0, T1, 8);  /* ABEF */
+  __m128i S1 = _mm_blend_epi16(T1, T0, 0xF0); /* CDGH */
+  while (nblk--) {
+    const __m128i AS = S0, CS = S1;
+    __m128i MSG, TMP;
+    __m128i M0 = _mm_shuffle_epi8(_mm_loadu_si128((const __m128i *)(data + 0)), MASK);
+    __m128i M1 = _mm_shuffle_epi8(_mm_loadu_si128((const __m128i *)(data + 16)), MASK);
+    __m128i M2 = _mm_shuffle_epi8(_mm_loadu_si128((const __m128i *)(data + 32)), MASK);
+    __m128i M3 = _mm_shuffle_epi8(_mm_loadu_si128((const __m128i *)(data + 48)), MASK);
+#define M3X_QROUND(M, k)                                                       \
+  MSG = _mm_add_epi32(M, _mm_loadu_si128((const __m128i *)&K[k]));             \
+  S1 = _mm_sha256rnds2_epu32(S1, S0, MSG);                                     \
+  MSG = _mm_shuffle_epi32(MSG, 0x0E);                                          \
+  S0 = _mm_sha256rnds2_epu32(S0, S1, MSG);
+#define M3X_SCHED(Ma, Mb, Mc, Md)                                              \
+  Ma = _mm_sha256msg1_epu32(Ma, Mb);                                           \
+  TMP = _mm_alignr_epi8(Md, Mc, 4);                                            \
+  Ma = _mm_add_epi32(Ma, TMP);                                                 \
+  Ma = _mm_sha256msg2_epu32(Ma, Md);
+    M3X_QROUND(M0, 0)
+    M3X_QROUND(M1, 4)
+    M3X_QROUND(M2, 8)
+    M3X_QROUND(M3, 12)
+    M3X_SCHED(M0, M1, M2, M3) M3X_QROUND(M0, 16)
+    M3X_SCHED(M1, M2, M3, M0) M3X_QROUND(M1, 20)
+    M3X_SCHED(M2, M3, M0, M1) M3X_QROUND(M2, 24)
+    M3X_SCHED(M3, M0, M1, M2) M3X_QROUND(M3, 28)
+    M3X_SCHED(M0, M1, M2, M3) M3X_QROUND(M0, 32)
+    M3X_SCHED(M1, M2, M3, M0) M3X_QROUND(M1, 36)
+    M3X_SCHED(M2, M3, M0, M1) M3X_QROUND(M2, 40)
+    M3X_SCHED(M3, M0, M1, M2) M3X_QROUND(M3, 44)
+    M3X_SCHED(M0, M1, M2, M3) M3X_QROUND(M0, 48)
+    M3X_SCHED(M1, M2, M3, M0) M3X_QROUND(M1, 52)
+    M3X_SCHED(M2, M3, M0, M1) M3X_QROUND(M2, 56)
+    M3X_SCHED(M3, M0, M1, M2) M3X_QROUND(M3, 60)
+#undef M3X_QROUND
+#undef M3X_SCHED
+    S0 = _mm_add_epi32(S0, AS);
+    S1 = _mm_add_epi32(S1, CS);
+    data += 64;
+  }
+  T0 = _mm_shuffle_epi32(S0, 0x1B);           /* FEBA */
+  T1 = _mm_shuffle_epi32(S1, 0xB1);           /* DCHG */
+  S0 = _mm_blend_epi16(T0, T1, 0xF0);         /* DCBA */
+  S1 = _mm_alignr_epi8(T1, T0, 8);            /* HGFE */
+  _mm_storeu_si128((__m128i *)st, S0);
+  _mm_storeu_si128((__m128i *)(st + 4), S1);
+}
+
+static int sha_ni_ok = -1;
+static int have_sha_ni(void) {
+  if (sha_ni_ok < 0) sha_ni_ok = __builtin_cpu_supports("sha") ? 1 : 0;
+  return sha_ni_ok;
+}
+#else
+static int have_sha_ni(void) { return 0; }
+#endif
+
+int m3x_oracle_have_sha_ni(void) { return have_sha_ni(); }
+
+static void compress_blocks(uint32_t st[8], const uint8_t *data, size_t nblk) {
+#if defined(__x86_64__)
+  if (have_sha_ni()) {
+    compress_ni(st, data, nblk);
+    return;
+  }
+#endif
+  while (nblk--) {
+    compress(st, data);
+    data += 64;
+  }
 }
