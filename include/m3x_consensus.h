@@ -140,6 +140,19 @@ int32_t m3x_shuffle_list_dev(m3x_ctx *ctx, void *indices_dev,
 
 /* key_validate a batch of compressed pubkeys: decompress + infinity reject +
  * subgroup check; uncomp[i] valid iff status[i]==0. (pubkey cache build) */
+/* RFC 9380 test entries (external-vector pinning; not on the hot path):
+ * the same device expand/SSWU/isogeny/cofactor code as the verify
+ * pipeline, parameterized by DST so the literal RFC appendix vectors
+ * (QUUX DSTs) apply. out_uniform (256B, optional NULL) exposes the
+ * expand_message_xmd output for the G2 suite's len_in_bytes=256. */
+int32_t m3x_bls_expand_test(m3x_ctx *ctx, const uint8_t *msg,
+                            uint32_t msg_len, const uint8_t *dst,
+                            uint32_t dst_len, uint32_t len_in_bytes,
+                            uint8_t *out);
+int32_t m3x_bls_h2c_test(m3x_ctx *ctx, const uint8_t *msg, uint32_t msg_len,
+                         const uint8_t *dst, uint32_t dst_len,
+                         uint8_t out_uncomp[192], uint8_t out_uniform[256]);
+
 int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp /* n*48 */,
                               uint64_t n, uint8_t *uncomp /* n*96 */,
                               int32_t *status /* n */);

@@ -157,6 +157,30 @@ def _bind(lib):
             ],
         ),
         (
+            "m3x_bls_expand_test",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+                ctypes.c_uint32,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
+            "m3x_bls_h2c_test",
+            [
+                ctypes.c_void_p,
+                ctypes.c_char_p,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+                ctypes.c_uint32,
+                ctypes.c_char_p,
+                ctypes.c_char_p,
+            ],
+        ),
+        (
             "m3x_bls_pk_decompress",
             [
                 ctypes.c_void_p,

@@ -1807,6 +1807,46 @@ __device__ inline void expand_message_xmd32(const uint8_t *msg,
   }
 }
 
+// General-DST expand_message_xmd (RFC 9380 §5.3.1, SHA-256): test-entry
+// form for the external RFC vectors (tests/golden/rfc9380_vectors.json);
+// the hot path keeps the fixed-DST expand_message_xmd32 below.
+__device__ inline void expand_message_xmd_gen(const uint8_t *msg,
+                                              uint32_t msg_len,
+                                              const uint8_t *dst,
+                                              uint32_t dst_len,
+                                              uint32_t len_in_bytes,
+                                              uint8_t *out) {
+  uint32_t ell = (len_in_bytes + 31) / 32;
+  uint8_t buf[64 + 544 + 2 + 1 + 255 + 1];
+  uint32_t off = 0;
+  for (int i = 0; i < 64; i++) buf[off++] = 0;
+  for (uint32_t i = 0; i < msg_len; i++) buf[off++] = msg[i];
+  buf[off++] = (uint8_t)(len_in_bytes >> 8);
+  buf[off++] = (uint8_t)len_in_bytes;
+  buf[off++] = 0;
+  for (uint32_t i = 0; i < dst_len; i++) buf[off++] = dst[i];
+  buf[off++] = (uint8_t)dst_len;
+  uint8_t b0[32];
+  m3x::sha256_bytes(buf, off, b0);
+  uint8_t cur[32 + 1 + 255 + 1];
+  for (int i = 0; i < 32; i++) cur[i] = b0[i];
+  cur[32] = 1;
+  for (uint32_t i = 0; i < dst_len; i++) cur[33 + i] = dst[i];
+  cur[33 + dst_len] = (uint8_t)dst_len;
+  uint8_t bi[32];
+  m3x::sha256_bytes(cur, 33 + dst_len + 1, bi);
+  uint32_t copied = len_in_bytes < 32 ? len_in_bytes : 32;
+  for (uint32_t i = 0; i < copied; i++) out[i] = bi[i];
+  for (uint32_t blk = 2; blk <= ell; blk++) {
+    for (int j = 0; j < 32; j++) cur[j] = b0[j] ^ bi[j];
+    cur[32] = (uint8_t)blk;
+    m3x::sha256_bytes(cur, 33 + dst_len + 1, bi);
+    uint32_t base = 32 * (blk - 1);
+    uint32_t nc = len_in_bytes - base < 32 ? len_in_bytes - base : 32;
+    for (uint32_t i = 0; i < nc; i++) out[base + i] = bi[i];
+  }
+}
+
 // SSWU tail given 1 + 1/tv already in hand (or the tv=0 constant case
 // signaled by tv_zero): shared by the single and dual entry points.
 __device__ inline void sswu_g2_tail(g2a &out, const fp2 &u, const fp2 &zu2,
@@ -2060,9 +2100,7 @@ __device__ inline void clear_cofactor_g2j(g2j &out, const g2j &p) {
   out = acc;
 }
 
-__device__ inline void h2c_g2(g2j &r, const uint8_t *msg) {
-  uint8_t uni[256];
-  expand_message_xmd32(msg, uni);
+__device__ inline void h2c_g2_from_uniform(g2j &r, const uint8_t uni[256]) {
   fp2 u[2];
   h2f_from_be64(u[0].c0, uni);
   h2f_from_be64(u[0].c1, uni + 64);
@@ -2076,6 +2114,20 @@ __device__ inline void h2c_g2(g2j &r, const uint8_t *msg) {
   g2j_add(s, s, t);
   clear_cofactor_g2j(r, s); // stays Jacobian: consumers (Q-Jacobian
                             // Miller loops) need no inversion at all
+}
+
+__device__ inline void h2c_g2(g2j &r, const uint8_t *msg) {
+  uint8_t uni[256];
+  expand_message_xmd32(msg, uni);
+  h2c_g2_from_uniform(r, uni);
+}
+
+// affine G2 -> 192B uncompressed wire form (x.c1||x.c0||y.c1||y.c0 BE)
+__device__ inline void g2_to_uncomp_dev(const g2a &p, uint8_t *out) {
+  fp_to_be48(p.x.c1, out);
+  fp_to_be48(p.x.c0, out + 48);
+  fp_to_be48(p.y.c1, out + 96);
+  fp_to_be48(p.y.c0, out + 144);
 }
 
 // G1 generator (negated y variant computed by callers when needed)

@@ -379,6 +379,35 @@ __global__ __launch_bounds__(64, 1) void k_bls_sig_aggregate_w(
   }
 }
 
+// one-thread test kernel: RFC 9380 h2c with an arbitrary DST, affine
+// uncompressed output — pins the device sswu/iso/cofactor code against
+// the literal RFC vectors (tests/golden/rfc9380_vectors.json)
+__global__ void k_bls_h2c_dst(const uint8_t *__restrict__ msg,
+                              uint32_t msg_len,
+                              const uint8_t *__restrict__ dst,
+                              uint32_t dst_len, uint8_t *__restrict__ out,
+                              uint8_t *__restrict__ uni_out) {
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  uint8_t uni[256];
+  expand_message_xmd_gen(msg, msg_len, dst, dst_len, 256, uni);
+  for (int i = 0; i < 256; i++) uni_out[i] = uni[i];
+  g2j h;
+  h2c_g2_from_uniform(h, uni);
+  g2a a;
+  g2j_to_aff(a, h);
+  g2_to_uncomp_dev(a, out);
+}
+
+// one-thread test kernel: general expand_message_xmd only
+__global__ void k_bls_expand_dst(const uint8_t *__restrict__ msg,
+                                 uint32_t msg_len,
+                                 const uint8_t *__restrict__ dst,
+                                 uint32_t dst_len, uint32_t len_in_bytes,
+                                 uint8_t *__restrict__ out) {
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  expand_message_xmd_gen(msg, msg_len, dst, dst_len, len_in_bytes, out);
+}
+
 int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                const void *pks_dev, const void *offs_dev,
                const void *rands_dev, uint64_t n, int32_t *out) {
@@ -485,22 +514,99 @@ int32_t m3x_bls_pk_decompress(m3x_ctx *ctx, const uint8_t *comp, uint64_t n,
   if (!ctx || n == 0) return M3X_ERR_ARG;
   std::lock_guard<std::recursive_mutex> lk(ctx->mu);
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
-  uint8_t *comp_d, *unc_d;
-  int32_t *st_d;
-  M3X_HIP_CHECK(hipMalloc(&comp_d, n * 48));
-  M3X_HIP_CHECK(hipMalloc(&unc_d, n * 96));
-  M3X_HIP_CHECK(hipMalloc(&st_d, n * 4));
-  hipMemcpyAsync(comp_d, comp, n * 48, hipMemcpyHostToDevice, ctx->stream);
-  uint32_t blocks = (uint32_t)((n + 63) / 64);
-  hipLaunchKernelGGL(k_bls_pk_decompress, dim3(blocks), dim3(64), 0,
-                     ctx->stream, comp_d, n, unc_d, st_d);
-  hipMemcpyAsync(uncomp, unc_d, n * 96, hipMemcpyDeviceToHost, ctx->stream);
-  hipMemcpyAsync(status, st_d, n * 4, hipMemcpyDeviceToHost, ctx->stream);
-  hipError_t e = hipStreamSynchronize(ctx->stream);
-  (void)hipFree(comp_d);
-  (void)hipFree(unc_d);
-  (void)hipFree(st_d);
-  return e == hipSuccess ? M3X_OK : M3X_ERR_HIP;
+  // single cleanup path: partial allocation failure frees prior buffers,
+  // and every copy is checked (a failed upload must not run the kernel on
+  // stale device memory and report "invalid")
+  uint8_t *comp_d = nullptr, *unc_d = nullptr;
+  int32_t *st_d = nullptr;
+  int32_t rc = M3X_ERR_HIP;
+  if (hipMalloc(&comp_d, n * 48) != hipSuccess) goto out;
+  if (hipMalloc(&unc_d, n * 96) != hipSuccess) goto out;
+  if (hipMalloc(&st_d, n * 4) != hipSuccess) goto out;
+  if (hipMemcpyAsync(comp_d, comp, n * 48, hipMemcpyHostToDevice,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  {
+    uint32_t blocks = (uint32_t)((n + 63) / 64);
+    hipLaunchKernelGGL(k_bls_pk_decompress, dim3(blocks), dim3(64), 0,
+                       ctx->stream, comp_d, n, unc_d, st_d);
+  }
+  if (hipMemcpyAsync(uncomp, unc_d, n * 96, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (hipMemcpyAsync(status, st_d, n * 4, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (hipStreamSynchronize(ctx->stream) == hipSuccess) rc = M3X_OK;
+out:
+  if (comp_d) (void)hipFree(comp_d);
+  if (unc_d) (void)hipFree(unc_d);
+  if (st_d) (void)hipFree(st_d);
+  return rc;
+}
+
+int32_t m3x_bls_expand_test(m3x_ctx *ctx, const uint8_t *msg,
+                            uint32_t msg_len, const uint8_t *dst,
+                            uint32_t dst_len, uint32_t len_in_bytes,
+                            uint8_t *out) {
+  if (!ctx || msg_len > 544 || dst_len == 0 || dst_len > 255 ||
+      len_in_bytes > 256)
+    return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *buf_d = nullptr;
+  int32_t rc = M3X_ERR_HIP;
+  if (hipMalloc(&buf_d, 1024 + 256) != hipSuccess) return M3X_ERR_HIP;
+  if (msg_len &&
+      hipMemcpyAsync(buf_d, msg, msg_len, hipMemcpyHostToDevice,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (hipMemcpyAsync(buf_d + 768, dst, dst_len, hipMemcpyHostToDevice,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  hipLaunchKernelGGL(k_bls_expand_dst, dim3(1), dim3(1), 0, ctx->stream,
+                     buf_d, msg_len, buf_d + 768, dst_len, len_in_bytes,
+                     buf_d + 1024);
+  if (hipMemcpyAsync(out, buf_d + 1024, len_in_bytes, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (hipStreamSynchronize(ctx->stream) == hipSuccess) rc = M3X_OK;
+out:
+  (void)hipFree(buf_d);
+  return rc;
+}
+
+int32_t m3x_bls_h2c_test(m3x_ctx *ctx, const uint8_t *msg, uint32_t msg_len,
+                         const uint8_t *dst, uint32_t dst_len,
+                         uint8_t out_uncomp[192], uint8_t out_uniform[256]) {
+  if (!ctx || msg_len > 544 || dst_len == 0 || dst_len > 255)
+    return M3X_ERR_ARG;
+  std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+  M3X_HIP_CHECK(hipSetDevice(ctx->device));
+  uint8_t *buf_d = nullptr;
+  int32_t rc = M3X_ERR_HIP;
+  if (hipMalloc(&buf_d, 1024 + 192 + 256) != hipSuccess) return M3X_ERR_HIP;
+  if (msg_len &&
+      hipMemcpyAsync(buf_d, msg, msg_len, hipMemcpyHostToDevice,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (hipMemcpyAsync(buf_d + 768, dst, dst_len, hipMemcpyHostToDevice,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  hipLaunchKernelGGL(k_bls_h2c_dst, dim3(1), dim3(1), 0, ctx->stream, buf_d,
+                     msg_len, buf_d + 768, dst_len, buf_d + 1024,
+                     buf_d + 1024 + 192);
+  if (hipMemcpyAsync(out_uncomp, buf_d + 1024, 192, hipMemcpyDeviceToHost,
+                     ctx->stream) != hipSuccess)
+    goto out;
+  if (out_uniform &&
+      hipMemcpyAsync(out_uniform, buf_d + 1024 + 192, 256,
+                     hipMemcpyDeviceToHost, ctx->stream) != hipSuccess)
+    goto out;
+  if (hipStreamSynchronize(ctx->stream) == hipSuccess) rc = M3X_OK;
+out:
+  (void)hipFree(buf_d);
+  return rc;
 }
 
 int32_t m3x_bls_sig_aggregate(m3x_ctx *ctx, const uint8_t *sigs, uint64_t n,
@@ -552,26 +658,37 @@ int32_t m3x_bls_verify_sets(m3x_ctx *ctx, const uint8_t *msgs,
   if (n == 0) return 0;
   M3X_HIP_CHECK(hipSetDevice(ctx->device));
   uint64_t nk = pk_offsets[n];
-  uint8_t *msgs_d, *sigs_d, *pks_d;
-  uint32_t *offs_d;
-  uint64_t *rands_d;
-  M3X_HIP_CHECK(hipMalloc(&msgs_d, n * 32));
-  M3X_HIP_CHECK(hipMalloc(&sigs_d, n * 96));
-  M3X_HIP_CHECK(hipMalloc(&pks_d, nk * 96));
-  M3X_HIP_CHECK(hipMalloc(&offs_d, (n + 1) * 4));
-  M3X_HIP_CHECK(hipMalloc(&rands_d, n * 8));
-  hipMemcpy(msgs_d, msgs, n * 32, hipMemcpyHostToDevice);
-  hipMemcpy(sigs_d, sigs, n * 96, hipMemcpyHostToDevice);
-  hipMemcpy(pks_d, pks, nk * 96, hipMemcpyHostToDevice);
-  hipMemcpy(offs_d, pk_offsets, (n + 1) * 4, hipMemcpyHostToDevice);
-  hipMemcpy(rands_d, rands, n * 8, hipMemcpyHostToDevice);
-  int32_t rc = m3x_bls_verify_sets_dev(ctx, msgs_d, sigs_d, pks_d, offs_d,
-                                       rands_d, n);
-  (void)hipFree(msgs_d);
-  (void)hipFree(sigs_d);
-  (void)hipFree(pks_d);
-  (void)hipFree(offs_d);
-  (void)hipFree(rands_d);
+  // checked uploads + single cleanup path: an infrastructure failure must
+  // surface as M3X_ERR_HIP, never as a "signature invalid" verdict from a
+  // kernel run over stale device memory
+  uint8_t *msgs_d = nullptr, *sigs_d = nullptr, *pks_d = nullptr;
+  uint32_t *offs_d = nullptr;
+  uint64_t *rands_d = nullptr;
+  int32_t rc = M3X_ERR_HIP;
+  if (hipMalloc(&msgs_d, n * 32) != hipSuccess) goto out;
+  if (hipMalloc(&sigs_d, n * 96) != hipSuccess) goto out;
+  if (hipMalloc(&pks_d, nk * 96) != hipSuccess) goto out;
+  if (hipMalloc(&offs_d, (n + 1) * 4) != hipSuccess) goto out;
+  if (hipMalloc(&rands_d, n * 8) != hipSuccess) goto out;
+  if (hipMemcpy(msgs_d, msgs, n * 32, hipMemcpyHostToDevice) != hipSuccess)
+    goto out;
+  if (hipMemcpy(sigs_d, sigs, n * 96, hipMemcpyHostToDevice) != hipSuccess)
+    goto out;
+  if (hipMemcpy(pks_d, pks, nk * 96, hipMemcpyHostToDevice) != hipSuccess)
+    goto out;
+  if (hipMemcpy(offs_d, pk_offsets, (n + 1) * 4, hipMemcpyHostToDevice) !=
+      hipSuccess)
+    goto out;
+  if (hipMemcpy(rands_d, rands, n * 8, hipMemcpyHostToDevice) != hipSuccess)
+    goto out;
+  rc = m3x_bls_verify_sets_dev(ctx, msgs_d, sigs_d, pks_d, offs_d, rands_d,
+                               n);
+out:
+  if (msgs_d) (void)hipFree(msgs_d);
+  if (sigs_d) (void)hipFree(sigs_d);
+  if (pks_d) (void)hipFree(pks_d);
+  if (offs_d) (void)hipFree(offs_d);
+  if (rands_d) (void)hipFree(rands_d);
   return rc;
 }
 

@@ -426,10 +426,8 @@ int32_t m3x_merkleize_chunks_dev(m3x_ctx *ctx, const void *chunks_dev,
   } else {
     rc = reduce_to_root(ctx, (const uint8_t *)chunks_dev, n_chunks, depth,
                         root_dev + 32);
-    if (rc != M3X_OK) {
-      (void)hipFree(root_dev);
-      return rc;
-    }
+    // root_dev points into ctx->small_pool — nothing to free on error
+    if (rc != M3X_OK) return rc;
     hipLaunchKernelGGL(k_finalize, dim3(1), dim3(64), 0, ctx->stream,
                        root_dev + 32, depth, depth, mix_len, ctx->zeros_dev,
                        root_dev);

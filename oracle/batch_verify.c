@@ -289,3 +289,47 @@ int m3x_oracle_bls_g1_mul(const uint8_t p_uncomp[96],
   g1_to_uncomp(&a, out);
   return 0;
 }
+
+int m3x_oracle_g2_subgroup_check(const uint8_t uncomp[192], int use_ref) {
+  /* introspection for tests: order-r membership of an uncompressed
+   * E'(Fp2) point, via the fast psi criterion (use_ref=0) or the
+   * [r]Q reference form (use_ref=1). Returns 1 in-subgroup, 0 not,
+   * <0 malformed/off-curve. */
+  bls_init();
+  g2_aff_t p;
+  if (g2_from_uncomp(&p, uncomp) != 0) return -1;
+  return use_ref ? g2_in_subgroup_ref(&p) : g2_in_subgroup(&p);
+}
+
+int m3x_oracle_bls_selftest(void) {
+  /* cross-checks of the round-2 fast paths against the reference forms
+   * (returns 0 ok, <0 which check failed):
+   *  1. final_exp3(f) == final_exp(f)^3
+   *  2. final_exp(miller_jacobian) == final_exp(miller_affine_ref)
+   *  3. psi subgroup check agrees with [r]Q on valid points */
+  bls_init();
+  uint8_t msg[32];
+  for (int i = 0; i < 32; i++) msg[i] = (uint8_t)(0x33 + i);
+  g2_aff_t h;
+  h2c_g2(&h, msg);
+  fp12_t f, fr;
+  fp12_one(&f);
+  miller(&f, &G1_GEN, &h);
+  fp12_one(&fr);
+  miller_affine_ref(&fr, &G1_GEN, &h);
+  fp12_t e1, e2, e3, t;
+  final_exp(&e1, &f);
+  final_exp(&e2, &fr);
+  /* 2: jacobian vs affine miller agree after final exp */
+  if (memcmp(&e1, &e2, sizeof(e1)) != 0) return -2;
+  /* 1: cubed chain == standard cubed */
+  final_exp3(&e3, &f);
+  fp12_mul_(&t, &e1, &e1);
+  fp12_mul_(&t, &t, &e1);
+  if (memcmp(&e3, &t, sizeof(t)) != 0) return -1;
+  /* 3: subgroup-check agreement on a valid point */
+  if (g2_in_subgroup(&h) != 1 || g2_in_subgroup_ref(&h) != 1) return -3;
+  if (g2_in_subgroup(&G2_GEN) != 1 || g2_in_subgroup_ref(&G2_GEN) != 1)
+    return -3;
+  return 0;
+}

@@ -1508,6 +1508,105 @@ void h2c_g2(g2_aff_t *r, const uint8_t msg[32]) {
   g2_to_aff(r, &cleared); /* single affine conversion */
 }
 
+/* ---- RFC 9380 general forms (arbitrary DST / msg len; external-vector
+ * pinning — the hot path keeps the fixed-DST fast forms above) ---- */
+
+void m3x_oracle_expand_xmd(const uint8_t *msg, uint32_t msg_len,
+                           const uint8_t *dst, uint32_t dst_len,
+                           uint32_t len_in_bytes, uint8_t *out) {
+  /* RFC 9380 §5.3.1, SHA-256 (b=32, s=64). Caller guarantees
+   * len_in_bytes <= 8160 and dst_len <= 255. */
+  uint32_t ell = (len_in_bytes + 31) / 32;
+  uint8_t buf[64 + 1024 + 2 + 1 + 255 + 1];
+  uint32_t off = 0;
+  memset(buf, 0, 64);
+  off = 64;
+  memcpy(buf + off, msg, msg_len);
+  off += msg_len;
+  buf[off++] = (uint8_t)(len_in_bytes >> 8);
+  buf[off++] = (uint8_t)len_in_bytes;
+  buf[off++] = 0;
+  memcpy(buf + off, dst, dst_len);
+  off += dst_len;
+  buf[off++] = (uint8_t)dst_len;
+  uint8_t b0[32];
+  m3x_oracle_sha256(buf, off, b0);
+  uint8_t cur[32 + 1 + 255 + 1];
+  memcpy(cur, b0, 32);
+  cur[32] = 1;
+  memcpy(cur + 33, dst, dst_len);
+  cur[33 + dst_len] = (uint8_t)dst_len;
+  uint8_t bi[32];
+  m3x_oracle_sha256(cur, 33 + dst_len + 1, bi);
+  uint32_t copied = len_in_bytes < 32 ? len_in_bytes : 32;
+  memcpy(out, bi, copied);
+  for (uint32_t blk = 2; blk <= ell; blk++) {
+    for (int j = 0; j < 32; j++) cur[j] = b0[j] ^ bi[j];
+    cur[32] = (uint8_t)blk;
+    m3x_oracle_sha256(cur, 33 + dst_len + 1, bi);
+    uint32_t base = 32 * (blk - 1);
+    uint32_t nc = len_in_bytes - base < 32 ? len_in_bytes - base : 32;
+    memcpy(out + base, bi, nc);
+  }
+}
+
+int m3x_oracle_h2c_g2_dst(const uint8_t *msg, uint32_t msg_len,
+                          const uint8_t *dst, uint32_t dst_len,
+                          uint8_t out_uncomp[192]) {
+  /* hash_to_curve for BLS12381G2_XMD:SHA-256_SSWU_RO_ with an arbitrary
+   * DST (RFC 9380 §8.8.2) — same sswu/iso/cofactor code as h2c_g2. */
+  bls_init();
+  uint8_t uni[256];
+  m3x_oracle_expand_xmd(msg, msg_len, dst, dst_len, 256, uni);
+  fp2_t u0, u1;
+  fp_from_be64_mod(&u0.c0, uni);
+  fp_from_be64_mod(&u0.c1, uni + 64);
+  fp_from_be64_mod(&u1.c0, uni + 128);
+  fp_from_be64_mod(&u1.c1, uni + 192);
+  g2_aff_t q0p, q1p, q0, q1;
+  sswu_g2(&q0p, &u0);
+  sswu_g2(&q1p, &u1);
+  iso_map_g2(&q0, &q0p);
+  iso_map_g2(&q1, &q1p);
+  g2_jac_t sj, tj;
+  g2_from_aff(&sj, &q0);
+  g2_from_aff(&tj, &q1);
+  g2_addj(&sj, &sj, &tj);
+  g2_jac_t cleared;
+  clear_cofactor_g2_jac(&cleared, &sj);
+  g2_aff_t r;
+  g2_to_aff(&r, &cleared);
+  g2_to_uncomp(&r, out_uncomp);
+  return 0;
+}
+
+/* map_to_curve WITHOUT cofactor clearing: yields an on-curve E'(Fp2)
+ * point outside G2 w.h.p. — negative-subgroup test material */
+int m3x_oracle_map_to_curve_g2_nococlear(const uint8_t msg[32],
+                                         uint8_t out_uncomp[192]) {
+  bls_init();
+  uint8_t uni[256];
+  expand_message_xmd(msg, 32, uni);
+  fp2_t u0, u1;
+  fp_from_be64_mod(&u0.c0, uni);
+  fp_from_be64_mod(&u0.c1, uni + 64);
+  fp_from_be64_mod(&u1.c0, uni + 128);
+  fp_from_be64_mod(&u1.c1, uni + 192);
+  g2_aff_t q0p, q1p, q0, q1;
+  sswu_g2(&q0p, &u0);
+  sswu_g2(&q1p, &u1);
+  iso_map_g2(&q0, &q0p);
+  iso_map_g2(&q1, &q1p);
+  g2_jac_t sj, tj;
+  g2_from_aff(&sj, &q0);
+  g2_from_aff(&tj, &q1);
+  g2_addj(&sj, &sj, &tj);
+  g2_aff_t r;
+  g2_to_aff(&r, &sj);
+  g2_to_uncomp(&r, out_uncomp);
+  return 0;
+}
+
 /* ----------------------------------------------------------------- init --- */
 
 static int bls_init_done = 0;

@@ -69,6 +69,19 @@ int m3x_oracle_bls_h2c_g2(const uint8_t msg[32], uint8_t out_uncomp[192]);
  * coefficients c0..c5 of Fp2[w]/(w^6-xi), each c0||c1). */
 int m3x_oracle_bls_pairing(const uint8_t p_uncomp[96],
                            const uint8_t q_uncomp[192], uint8_t out[576]);
+/* RFC 9380 general forms (external-vector pinning; tests only) */
+void m3x_oracle_expand_xmd(const uint8_t *msg, uint32_t msg_len,
+                           const uint8_t *dst, uint32_t dst_len,
+                           uint32_t len_in_bytes, uint8_t *out);
+int m3x_oracle_h2c_g2_dst(const uint8_t *msg, uint32_t msg_len,
+                          const uint8_t *dst, uint32_t dst_len,
+                          uint8_t out_uncomp[192]);
+int m3x_oracle_map_to_curve_g2_nococlear(const uint8_t msg[32],
+                                         uint8_t out_uncomp[192]);
+
+int m3x_oracle_g2_subgroup_check(const uint8_t uncomp[192], int use_ref);
+int m3x_oracle_bls_selftest(void);
+
 /* [scalar_be32]P for G1 (96B uncomp in/out); for tests. */
 int m3x_oracle_bls_g1_mul(const uint8_t p_uncomp[96],
                           const uint8_t scalar_be[32], uint8_t out[96]);

@@ -1,0 +1,215 @@
+"""RFC 9380 external pinning (VERDICT round-1 item 2).
+
+tests/golden/rfc9380_vectors.json embeds literal RFC 9380 appendix
+vectors: K.1 expand_message_xmd (SHA-256, QUUX expander DST) and J.10.1
+BLS12381G2_XMD:SHA-256_SSWU_RO_ hash_to_curve (QUUX suite DST). Both the
+CPU oracle and the HIP kernels must reproduce them — this pins the whole
+h2c pipeline (expand → SSWU → 3-isogeny → cofactor clearing) to public
+spec data, independent of the in-repo generation script.
+
+Also the negative-subgroup case (ADVICE round-1): an on-curve E'(Fp2)
+point built WITHOUT cofactor clearing must be rejected as a signature by
+the psi-based subgroup check (oracle fast + [r]Q reference + GPU).
+"""
+import ctypes
+import json
+from pathlib import Path
+
+import pytest
+
+GOLDEN = json.loads(
+    (Path(__file__).parent / "golden" / "rfc9380_vectors.json").read_text()
+)
+
+P_MOD = int(
+    "1a0111ea397fe69a4b1ba7b6434bacd764774b84f38512bf6730d2a0f6b0f624"
+    "1eabfffeb153ffffb9feffffffffaaab",
+    16,
+)
+
+
+def _split_uncomp(raw):
+    """192B uncompressed G2 (x.c1||x.c0||y.c1||y.c0 BE) -> hex dict."""
+    h = raw.hex()
+    return {
+        "P_x_c1": h[0:96],
+        "P_x_c0": h[96:192],
+        "P_y_c1": h[192:288],
+        "P_y_c0": h[288:384],
+    }
+
+
+def _compress_g2(uncomp):
+    """96B compressed form of an uncompressed G2 point (ZCash flags)."""
+    x_c1 = uncomp[:48]
+    y_c1 = int.from_bytes(uncomp[96:144], "big")
+    y_c0 = int.from_bytes(uncomp[144:192], "big")
+    half = (P_MOD - 1) // 2
+    if y_c1 != 0:
+        big = y_c1 > half
+    else:
+        big = y_c0 > half
+    out = bytearray(uncomp[:96])  # x.c1 || x.c0
+    out[0] |= 0x80
+    if big:
+        out[0] |= 0x20
+    return bytes(out)
+
+
+# ---------------------------------------------------------------- oracle ---
+
+
+def test_oracle_expand_xmd_rfc_vectors(oracle):
+    sec = GOLDEN["expand_xmd_sha256"]
+    dst = sec["dst"].encode()
+    for case in sec["cases"]:
+        msg = case["msg"].encode()
+        out = ctypes.create_string_buffer(case["len"])
+        oracle.m3x_oracle_expand_xmd(
+            msg, len(msg), dst, len(dst), case["len"], out
+        )
+        assert out.raw.hex() == case["uniform_bytes"], case["msg"][:16]
+
+
+def test_oracle_h2c_g2_rfc_vectors(oracle):
+    sec = GOLDEN["h2c_g2_sswu_ro"]
+    dst = sec["dst"].encode()
+    for case in sec["cases"]:
+        msg = case["msg"].encode()
+        out = ctypes.create_string_buffer(192)
+        rc = oracle.m3x_oracle_h2c_g2_dst(msg, len(msg), dst, len(dst), out)
+        assert rc == 0
+        got = _split_uncomp(out.raw)
+        for k in ("P_x_c0", "P_x_c1", "P_y_c0", "P_y_c1"):
+            assert got[k] == case[k], (case["msg"], k)
+
+
+def test_oracle_fast_path_selftest(oracle):
+    """Jacobian Miller == affine-ref Miller (after final exp), cubed
+    final-exp chain == standard^3, psi subgroup check == [r]Q check."""
+    assert oracle.m3x_oracle_bls_selftest() == 0
+
+
+def test_oracle_non_subgroup_point_rejected(oracle):
+    pt = bytes.fromhex(GOLDEN["non_subgroup_g2"]["uncompressed"])
+    # outside G2 by BOTH criteria; still a valid curve point (decompress
+    # of its compressed form recomputes y from the curve equation and
+    # must round-trip)
+    assert oracle.m3x_oracle_g2_subgroup_check(pt, 0) == 0
+    assert oracle.m3x_oracle_g2_subgroup_check(pt, 1) == 0
+    comp = _compress_g2(pt)
+    unc = ctypes.create_string_buffer(192)
+    assert oracle.m3x_oracle_bls_sig_decompress(comp, unc) == 0
+    assert unc.raw == pt
+    # and as a signature it must verify False (blst.rs:73-77 deferred
+    # subgroup check), not error
+    sks = ctypes.create_string_buffer(32)
+    pks = ctypes.create_string_buffer(96)
+    oracle.m3x_oracle_bls_keypool(1, sks, pks)
+    msg = bytes.fromhex(GOLDEN["non_subgroup_g2"]["msg"])
+    offs = (ctypes.c_uint32 * 2)(0, 1)
+    rnds = (ctypes.c_uint64 * 1)(7)
+    v = oracle.m3x_oracle_bls_verify_sets(msg, comp, pks.raw, offs, rnds, 1)
+    assert v == 0
+
+
+def test_oracle_torsion_valid_signature_still_passes(oracle):
+    """control for the negative test: a well-formed signature over the
+    same message verifies True."""
+    import hashlib
+
+    sks = ctypes.create_string_buffer(32)
+    pks = ctypes.create_string_buffer(96)
+    oracle.m3x_oracle_bls_keypool(1, sks, pks)
+    msg = hashlib.sha256(b"torsion-control").digest()
+    sig = ctypes.create_string_buffer(96)
+    assert oracle.m3x_oracle_bls_sign_batch(1, sks.raw, msg, sig) == 0
+    offs = (ctypes.c_uint32 * 2)(0, 1)
+    rnds = (ctypes.c_uint64 * 1)(7)
+    assert (
+        oracle.m3x_oracle_bls_verify_sets(msg, sig.raw, pks.raw, offs, rnds, 1)
+        == 1
+    )
+
+
+# ------------------------------------------------------------------- GPU ---
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    from lighthouse_amd import _native
+
+    return _native.default_ctx()
+
+
+@pytest.mark.gpu
+def test_gpu_expand_xmd_rfc_vectors(ctx):
+    sec = GOLDEN["expand_xmd_sha256"]
+    dst = sec["dst"].encode()
+    for case in sec["cases"]:
+        msg = case["msg"].encode()
+        out = ctypes.create_string_buffer(case["len"])
+        rc = ctx._lib.m3x_bls_expand_test(
+            ctx.handle, msg, len(msg), dst, len(dst), case["len"], out
+        )
+        assert rc == 0
+        assert out.raw.hex() == case["uniform_bytes"], case["msg"][:16]
+
+
+@pytest.mark.gpu
+def test_gpu_h2c_rfc_vectors(ctx):
+    """k_bls_h2c device code vs the literal RFC 9380 J.10.1 vectors."""
+    sec = GOLDEN["h2c_g2_sswu_ro"]
+    dst = sec["dst"].encode()
+    for case in sec["cases"]:
+        msg = case["msg"].encode()
+        out = ctypes.create_string_buffer(192)
+        uni = ctypes.create_string_buffer(256)
+        rc = ctx._lib.m3x_bls_h2c_test(
+            ctx.handle, msg, len(msg), dst, len(dst), out, uni
+        )
+        assert rc == 0
+        got = _split_uncomp(out.raw)
+        for k in ("P_x_c0", "P_x_c1", "P_y_c0", "P_y_c1"):
+            assert got[k] == case[k], (case["msg"], k)
+
+
+@pytest.mark.gpu
+def test_gpu_non_subgroup_signature_rejected(ctx):
+    """GPU psi-based subgroup check rejects an on-curve non-G2 signature
+    (and the control with a valid signature passes) — ADVICE round-1."""
+    from lighthouse_amd import bls
+
+    fx = json.loads(
+        (Path(__file__).parent / "golden" / "bls_fixtures.json").read_text()
+    )
+    interop = fx["interop"][0]
+    pk = bls.PublicKey.from_uncompressed(
+        bytes.fromhex(interop["pk_uncompressed_hex"])
+    )
+    pt = bytes.fromhex(GOLDEN["non_subgroup_g2"]["uncompressed"])
+    comp = _compress_g2(pt)
+    msg = bytes.fromhex(GOLDEN["non_subgroup_g2"]["msg"])
+    bad = bls.SignatureSet(
+        signature=bls.Signature.from_compressed(comp),
+        signing_keys=[pk],
+        message=msg,
+    )
+    assert bls.verify_signature_sets([bad], ctx=ctx, _rands=[7]) is False
+    # control: a valid fixture batch still verifies True on this ctx
+    case = next(c for c in fx["batch_cases"] if c["name"] == "ten_sets_valid")
+    sets = [
+        bls.SignatureSet(
+            signature=bls.Signature.from_compressed(
+                bytes.fromhex(s["sig_compressed_hex"])
+            ),
+            signing_keys=[
+                bls.PublicKey.from_uncompressed(bytes.fromhex(p))
+                for p in s["pks_uncompressed_hex"]
+            ],
+            message=bytes.fromhex(s["msg_hex"]),
+        )
+        for s in case["sets"]
+    ]
+    rands = [int(r) for r in case["rands"]]
+    assert bls.verify_signature_sets(sets, ctx=ctx, _rands=rands) is True
