@@ -327,11 +327,23 @@ class Ctx:
         return out.raw
 
 
-_default_ctx = None
+_tls = None
 
 
 def default_ctx() -> Ctx:
-    global _default_ctx
-    if _default_ctx is None:
-        _default_ctx = Ctx(int(os.environ.get("M3X_DEVICE", "0")))
-    return _default_ctx
+    """Per-THREAD context (SURVEY §8b threading contract: the backend is
+    called concurrently from N beacon-processor workers, so it must be
+    re-entrant and support concurrent batches). Each thread gets its own
+    context — own HIP streams + scratch — so concurrent
+    verify_signature_sets calls from different threads run on different
+    streams without serializing on one context's lock."""
+    global _tls
+    if _tls is None:
+        import threading
+
+        _tls = threading.local()
+    ctx = getattr(_tls, "ctx", None)
+    if ctx is None:
+        ctx = Ctx(int(os.environ.get("M3X_DEVICE", "0")))
+        _tls.ctx = ctx
+    return ctx

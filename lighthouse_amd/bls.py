@@ -202,11 +202,47 @@ def eth_fast_aggregate_verify(signature: "Signature", msg: bytes, pubkeys,
     return fast_aggregate_verify(signature, msg, pubkeys, ctx=ctx)
 
 
-# aggregate_verify (distinct messages under ONE aggregate signature,
-# blst.rs:263-274) exists in the reference "only for EF tests, presently
-# not used in production" (generic_aggregate_signature.rs:44-47); it is not
-# expressible as independent signature sets and is served by the oracle in
-# tests until a dedicated kernel is warranted.
+def aggregate_verify(signature: "Signature", msgs, pubkeys, ctx=None) -> bool:
+    """TAggregateSignature::aggregate_verify (blst.rs:263-274): ONE
+    aggregate signature over n distinct messages, one pubkey each —
+    e(g1, sig) == prod_i e(pk_i, H(m_i)).
+
+    Runs on the existing batch pipeline with no extra kernel: n sets
+    {msg_i, pk_i} with sigma_1 = sig, sigma_i>1 = infinity and r_i = 1
+    compute prod e(PK_i, H(m_i)) * e(-g1, sig) == 1 — exactly the
+    aggregate_verify equation. r=1 is correct here: this is a single
+    equation, not a batch of independent claims, and the reference's
+    aggregate_verify uses no randomizers either. (The reference keeps
+    this 'only for EF tests' — generic_aggregate_signature.rs:44-47.)"""
+    msgs = [bytes(m) for m in msgs]
+    pubkeys = list(pubkeys)
+    if not msgs or len(msgs) != len(pubkeys):
+        return False
+    if signature.is_empty():
+        return False
+    sets = [
+        SignatureSet(
+            signature if i == 0 else Signature.infinity(), [pk], m
+        )
+        for i, (m, pk) in enumerate(zip(msgs, pubkeys))
+    ]
+    return verify_signature_sets(sets, ctx=ctx, _rands=[1] * len(sets))
+
+
+def verify_signature_sets_with_fallback(sets, ctx=None):
+    """Gossip-batch driver semantics (attestation_verification/
+    batch.rs:109-127): one batch check; if the batch verdict is false,
+    re-verify each set individually to identify the bad items (the
+    reference maps each result back to accept/reject per attestation).
+    Returns a per-set list of bools."""
+    sets = list(sets)
+    if verify_signature_sets(sets, ctx=ctx):
+        return [True] * len(sets)
+    # set.verify() path (generic_signature_set.rs:111-120): single-set
+    # fast_aggregate_verify, no randomizer
+    return [
+        verify_signature_sets([s], ctx=ctx, _rands=[1]) for s in sets
+    ]
 
 
 def verify_signature_sets(sets, ctx=None, _rands=None) -> bool:
