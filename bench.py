@@ -409,8 +409,13 @@ def main():
     cpu_baseline = None
     cpu_sha = None
     if rank == 0 and world == 1 and not DRY:
-        # BLS: 256-set sample, same 75/25 mix (192 k=1 + 64 k=512)
-        idx = list(range(192)) + list(range(N_SETS - N_AGG, N_SETS - N_AGG + 64))
+        # BLS: 16,384-set sample, same 75/25 mix (12,288 k=1 + 4,096
+        # k=512) — large enough that 256 OpenMP threads are saturated and
+        # the per-batch final exponentiation amortizes (the round-1
+        # 256-set sample measured thread-spawn, not throughput)
+        idx = list(range(12288)) + list(
+            range(N_SETS - N_AGG, N_SETS - N_AGG + 4096)
+        )
         msgs = b"".join(bls_w["msgs"][32 * i : 32 * (i + 1)] for i in idx)
         sigs = b"".join(bls_w["sigs"][96 * i : 96 * (i + 1)] for i in idx)
         pkb = b""
@@ -421,19 +426,27 @@ def main():
             offs.append(offs[-1] + (b - a))
         off_arr = (ctypes.c_uint32 * len(offs))(*offs)
         rnd = (ctypes.c_uint64 * len(idx))(*[bls_w["rands"][i] for i in idx])
-        tb = time.time()
-        vcpu = oracle.m3x_oracle_bls_verify_sets(
-            msgs, sigs, pkb, off_arr, rnd, ctypes.c_uint64(len(idx))
-        )
-        tcpu = time.time() - tb
-        assert vcpu == 1
+        tcpu = None
+        for _ in range(2):  # warm thread pool, then measure (best of 2)
+            tb = time.time()
+            vcpu = oracle.m3x_oracle_bls_verify_sets(
+                msgs, sigs, pkb, off_arr, rnd, ctypes.c_uint64(len(idx))
+            )
+            t = time.time() - tb
+            assert vcpu == 1
+            tcpu = t if tcpu is None else min(tcpu, t)
         cores = os.cpu_count()
         cpu_baseline = {
             "value": len(idx) / tcpu,
             "unit": "sets/s",
             "cores": cores,
             "kind": "port",
-            "sample": f"{len(idx)} sets (75% k=1, 25% k=512), {tcpu:.1f}s, OpenMP x{cores}",
+            "sample": (
+                f"{len(idx)} sets (75% k=1, 25% k=512), {tcpu:.2f}s best-of-2,"
+                f" OpenMP x{cores} = {len(idx)/tcpu/cores:.0f} sets/s/core"
+                " (optimized-port oracle; blst-class asm is ~1000-1500"
+                " sets/s/core — see BASELINE.md)"
+            ),
         }
         # shuffle CPU baseline (oracle, single list)
         import numpy as _np

@@ -118,6 +118,106 @@ void fp_mul_(fp_t *r, const fp_t *a, const fp_t *b) {
 
 static void fp_sqr(fp_t *r, const fp_t *a) { fp_mul_(r, a, a); }
 
+/* ---- lazy-reduction support (round 2): 12-limb full product + one
+ * Montgomery reduction, so fp2_mul does 3 wide muls + 2 reductions
+ * instead of 3 full CIOS (≈17% fewer MACs; bounds checked below). ---- */
+
+typedef struct { uint64_t w[12]; } fp_wide_t;
+
+/* t = a*b, full 768-bit product (no reduction) */
+static void fp_mul_wide(fp_wide_t *t, const fp_t *a, const fp_t *b) {
+  uint64_t w0=0,w1=0,w2=0,w3=0,w4=0,w5=0,w6=0,w7=0,w8=0,w9=0,w10=0,w11=0;
+  const uint64_t *A = a->v, *B = b->v;
+  uint64_t c;
+#define M3X_MACW(hi, lo, x, y, add0, add1)                                     \
+  do {                                                                         \
+    unsigned __int128 _p = (unsigned __int128)(x) * (y) + (add0) + (add1);     \
+    (lo) = (uint64_t)_p;                                                       \
+    (hi) = (uint64_t)(_p >> 64);                                               \
+  } while (0)
+#define M3X_ROWW(i, r0, r1, r2, r3, r4, r5, r6)                                \
+  do {                                                                         \
+    M3X_MACW(c, r0, A[0], B[i], r0, 0);                                        \
+    M3X_MACW(c, r1, A[1], B[i], r1, c);                                        \
+    M3X_MACW(c, r2, A[2], B[i], r2, c);                                        \
+    M3X_MACW(c, r3, A[3], B[i], r3, c);                                        \
+    M3X_MACW(c, r4, A[4], B[i], r4, c);                                        \
+    M3X_MACW(c, r5, A[5], B[i], r5, c);                                        \
+    r6 = c;                                                                    \
+  } while (0)
+  M3X_ROWW(0, w0, w1, w2, w3, w4, w5, w6);
+  M3X_ROWW(1, w1, w2, w3, w4, w5, w6, w7);
+  M3X_ROWW(2, w2, w3, w4, w5, w6, w7, w8);
+  M3X_ROWW(3, w3, w4, w5, w6, w7, w8, w9);
+  M3X_ROWW(4, w4, w5, w6, w7, w8, w9, w10);
+  M3X_ROWW(5, w5, w6, w7, w8, w9, w10, w11);
+#undef M3X_ROWW
+  t->w[0]=w0; t->w[1]=w1; t->w[2]=w2; t->w[3]=w3; t->w[4]=w4; t->w[5]=w5;
+  t->w[6]=w6; t->w[7]=w7; t->w[8]=w8; t->w[9]=w9; t->w[10]=w10; t->w[11]=w11;
+}
+
+/* wide add/sub (no reduction); caller tracks bounds */
+static void fp_wide_add(fp_wide_t *r, const fp_wide_t *a, const fp_wide_t *b) {
+  unsigned __int128 cc = 0;
+  for (int i = 0; i < 12; i++) {
+    cc += (unsigned __int128)a->w[i] + b->w[i];
+    r->w[i] = (uint64_t)cc;
+    cc >>= 64;
+  }
+}
+
+/* r = a - b + 2p^2 (keeps the value positive; 2p^2 < 2^763) */
+static uint64_t PP2_[12]; /* 2*p^2, set in bls_init */
+static void fp_wide_sub_pp2(fp_wide_t *r, const fp_wide_t *a,
+                            const fp_wide_t *b) {
+  unsigned __int128 cc = 0;
+  uint64_t t[12];
+  for (int i = 0; i < 12; i++) {
+    cc += (unsigned __int128)a->w[i] + PP2_[i];
+    t[i] = (uint64_t)cc;
+    cc >>= 64;
+  }
+  unsigned __int128 bw = 0;
+  for (int i = 0; i < 12; i++) {
+    unsigned __int128 x = (unsigned __int128)t[i] - b->w[i] - (uint64_t)bw;
+    r->w[i] = (uint64_t)x;
+    bw = (x >> 64) & 1;
+  }
+}
+
+/* Montgomery reduction of a 12-limb T < p*2^384: r = T*R^-1 mod p */
+static void fp_redc(fp_t *r, const fp_wide_t *T) {
+  uint64_t t0=T->w[0],t1=T->w[1],t2=T->w[2],t3=T->w[3],t4=T->w[4],t5=T->w[5];
+  uint64_t hi6=T->w[6],hi7=T->w[7],hi8=T->w[8],hi9=T->w[9],hi10=T->w[10],hi11=T->w[11];
+  uint64_t carry6 = 0; /* accumulated carries into the high half */
+  uint64_t c, m;
+#define M3X_REDR()                                                             \
+  do {                                                                         \
+    m = t0 * BLS_N0;                                                           \
+    {                                                                          \
+      unsigned __int128 p0 = (unsigned __int128)m * BLS_P[0] + t0;             \
+      c = (uint64_t)(p0 >> 64);                                                \
+    }                                                                          \
+    M3X_MACW(c, t0, m, BLS_P[1], t1, c);                                       \
+    M3X_MACW(c, t1, m, BLS_P[2], t2, c);                                       \
+    M3X_MACW(c, t2, m, BLS_P[3], t3, c);                                       \
+    M3X_MACW(c, t3, m, BLS_P[4], t4, c);                                       \
+    M3X_MACW(c, t4, m, BLS_P[5], t5, c);                                       \
+  } while (0)
+  /* 6 rounds; after each, shift in the next high limb + carry */
+  unsigned __int128 s;
+  M3X_REDR(); s = (unsigned __int128)hi6 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+  M3X_REDR(); s = (unsigned __int128)hi7 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+  M3X_REDR(); s = (unsigned __int128)hi8 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+  M3X_REDR(); s = (unsigned __int128)hi9 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+  M3X_REDR(); s = (unsigned __int128)hi10 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+  M3X_REDR(); s = (unsigned __int128)hi11 + c + carry6; t5 = (uint64_t)s; carry6 = (uint64_t)(s >> 64);
+#undef M3X_REDR
+  uint64_t t[6] = {t0, t1, t2, t3, t4, t5};
+  if (carry6 || ge_p(t)) sub_p(t);
+  memcpy(r->v, t, 48);
+}
+
 int fp_is_zero_(const fp_t *a) {
   uint64_t o = 0;
   for (int i = 0; i < 6; i++) o |= a->v[i];
@@ -232,16 +332,22 @@ static void fp2_neg(fp2_t *r, const fp2_t *a) {
   fp_neg(&r->c1, &a->c1);
 }
 void fp2_mul_(fp2_t *r, const fp2_t *a, const fp2_t *b) {
-  fp_t t0, t1, s0, s1, m;
-  fp_mul_(&t0, &a->c0, &b->c0);
-  fp_mul_(&t1, &a->c1, &b->c1);
-  fp_add_(&s0, &a->c0, &a->c1);
-  fp_add_(&s1, &b->c0, &b->c1);
-  fp_mul_(&m, &s0, &s1);
-  fp_sub_(&m, &m, &t0);
-  fp_sub_(&m, &m, &t1);
-  fp_sub_(&r->c0, &t0, &t1);
-  memcpy(&r->c1, &m, sizeof(fp_t));
+  /* Karatsuba with LAZY reduction: 3 wide products + 2 Montgomery
+   * reductions (vs 3 full CIOS). Offsets by 2p^2 keep the wide
+   * differences positive; all inputs < p so every reduction input
+   * < 3p^2 < p*2^384 (single conditional subtract). */
+  fp_wide_t t0, t1, m, s01, cw;
+  fp_t sa, sb;
+  fp_mul_wide(&t0, &a->c0, &b->c0);
+  fp_mul_wide(&t1, &a->c1, &b->c1);
+  fp_add_(&sa, &a->c0, &a->c1);
+  fp_add_(&sb, &b->c0, &b->c1);
+  fp_mul_wide(&m, &sa, &sb);
+  fp_wide_sub_pp2(&cw, &t0, &t1); /* c0 = t0 - t1 */
+  fp_redc(&r->c0, &cw);
+  fp_wide_add(&s01, &t0, &t1);
+  fp_wide_sub_pp2(&cw, &m, &s01); /* c1 = m - t0 - t1 */
+  fp_redc(&r->c1, &cw);
 }
 static void fp2_sqr(fp2_t *r, const fp2_t *a) {
   fp_t s, d, m;
@@ -1609,10 +1715,32 @@ int m3x_oracle_map_to_curve_g2_nococlear(const uint8_t msg[32],
 
 /* ----------------------------------------------------------------- init --- */
 
+static void pp2_init(void) {
+  /* PP2_ = 2 * p^2 as a 768-bit integer (lazy-reduction offset) */
+  unsigned __int128 acc;
+  uint64_t tmp[13] = {0};
+  for (int i = 0; i < 6; i++) {
+    uint64_t carry = 0;
+    for (int j = 0; j < 6; j++) {
+      acc = (unsigned __int128)BLS_P[i] * BLS_P[j] + tmp[i + j] + carry;
+      tmp[i + j] = (uint64_t)acc;
+      carry = (uint64_t)(acc >> 64);
+    }
+    tmp[i + 6] += carry;
+  }
+  uint64_t c = 0;
+  for (int i = 0; i < 12; i++) {
+    uint64_t v = (tmp[i] << 1) | c;
+    c = tmp[i] >> 63;
+    PP2_[i] = v;
+  }
+}
+
 static int bls_init_done = 0;
 
 void bls_init(void) {
   if (bls_init_done) return;
+  pp2_init();
   /* Montgomery constants */
   memset(&FP_ZERO_, 0, sizeof(FP_ZERO_));
   memcpy(FP_R2_.v, BLS_R2, 48);
