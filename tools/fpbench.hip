@@ -303,6 +303,238 @@ __global__ void chainF(const uint64_t *in, uint64_t *out, int iters) {
   for (int i = 0; i < 6; i++) { o[i] = a1.v[i]; o[6 + i] = a2.v[i]; }
 }
 
+// ======================= round-2 variants =======================
+// PP2 = 2*p^2 as 24 u32 limbs (lazy-reduction offset), compile-time
+struct pp2_t { uint32_t w[24]; };
+constexpr pp2_t make_pp2() {
+  pp2_t r{};
+  uint64_t t[13] = {};
+  for (int i = 0; i < 6; i++) {
+    uint64_t carry = 0;
+    for (int j = 0; j < 6; j++) {
+      unsigned __int128 acc =
+          (unsigned __int128)BLS_P[i] * BLS_P[j] + t[i + j] + carry;
+      t[i + j] = (uint64_t)acc;
+      carry = (uint64_t)(acc >> 64);
+    }
+    // propagate into the (fresh-enough) tail
+    int k = i + 6;
+    while (carry) {
+      unsigned __int128 acc = (unsigned __int128)t[k] + carry;
+      t[k] = (uint64_t)acc;
+      carry = (uint64_t)(acc >> 64);
+      k++;
+    }
+  }
+  uint64_t c = 0;
+  for (int i = 0; i < 12; i++) {
+    uint64_t v = (t[i] << 1) | c;
+    c = t[i] >> 63;
+    r.w[2 * i] = (uint32_t)v;
+    r.w[2 * i + 1] = (uint32_t)(v >> 32);
+  }
+  return r;
+}
+__constant__ constexpr pp2_t PP2 = make_pp2();
+
+__device__ __forceinline__ void mul_wide32(uint32_t t[25], const fp &a,
+                                           const fp &b) {
+  const uint32_t *A = reinterpret_cast<const uint32_t *>(a.v);
+  const uint32_t *B = reinterpret_cast<const uint32_t *>(b.v);
+#pragma unroll
+  for (int i = 0; i < 25; i++) t[i] = 0;
+#pragma unroll
+  for (int i = 0; i < 12; i++) {
+    uint64_t c = 0;
+#pragma unroll
+    for (int j = 0; j < 12; j++) {
+      uint64_t s = (uint64_t)A[j] * B[i] + t[i + j] + (uint32_t)c;
+      t[i + j] = (uint32_t)s;
+      c = s >> 32;
+    }
+    t[i + 12] = (uint32_t)c;
+  }
+}
+
+// Montgomery reduce a 24-limb T < 3p^2 (< p*2^384): r = T*R^-1 mod p
+__device__ __forceinline__ void redc32(fp &r, const uint32_t *T) {
+  uint32_t P32[12];
+#pragma unroll
+  for (int i = 0; i < 6; i++) {
+    P32[2 * i] = (uint32_t)BLS_P[i];
+    P32[2 * i + 1] = (uint32_t)(BLS_P[i] >> 32);
+  }
+  const uint32_t N0_32 = (uint32_t)BLS_N0;
+  uint32_t t[12];
+#pragma unroll
+  for (int i = 0; i < 12; i++) t[i] = T[i];
+  uint32_t carry = 0;
+#pragma unroll
+  for (int i = 0; i < 12; i++) {
+    uint32_t m = t[0] * N0_32;
+    uint64_t c = ((uint64_t)m * P32[0] + t[0]) >> 32;
+#pragma unroll
+    for (int j = 1; j < 12; j++) {
+      uint64_t s = (uint64_t)m * P32[j] + t[j] + (uint32_t)c;
+      t[j - 1] = (uint32_t)s;
+      c = s >> 32;
+    }
+    uint64_t s = (uint64_t)T[12 + i] + c + carry;
+    t[11] = (uint32_t)s;
+    carry = (uint32_t)(s >> 32);
+  }
+  uint64_t res[6];
+#pragma unroll
+  for (int i = 0; i < 6; i++)
+    res[i] = (uint64_t)t[2 * i] | ((uint64_t)t[2 * i + 1] << 32);
+  if (carry || fp_ge_p(res)) fp_sub_p(res);
+#pragma unroll
+  for (int i = 0; i < 6; i++) r.v[i] = res[i];
+}
+
+// r = a + PP2 - b over 24 limbs (keeps the difference positive)
+__device__ __forceinline__ void wide_sub_pp2(uint32_t r[25],
+                                             const uint32_t a[25],
+                                             const uint32_t b[25]) {
+  uint64_t c = 0;
+  uint32_t tmp[24];
+#pragma unroll
+  for (int i = 0; i < 24; i++) {
+    c += (uint64_t)a[i] + PP2.w[i];
+    tmp[i] = (uint32_t)c;
+    c >>= 32;
+  }
+  int64_t bw = 0;
+#pragma unroll
+  for (int i = 0; i < 24; i++) {
+    int64_t x = (int64_t)(uint64_t)tmp[i] - b[i] - bw;
+    r[i] = (uint32_t)x;
+    bw = (x < 0);
+  }
+}
+
+__device__ __forceinline__ void wide_add(uint32_t r[25], const uint32_t a[25],
+                                         const uint32_t b[25]) {
+  uint64_t c = 0;
+#pragma unroll
+  for (int i = 0; i < 24; i++) {
+    c += (uint64_t)a[i] + b[i];
+    r[i] = (uint32_t)c;
+    c >>= 32;
+  }
+}
+
+// lazy-reduction Karatsuba fp2 mul: 3 wide products + 2 reductions
+__device__ __forceinline__ void mul2_lazy(fp2 &r, const fp2 &a,
+                                          const fp2 &b) {
+  uint32_t t0[25], t1[25], m[25], s01[25], cw[25];
+  fp sa, sb;
+  mul_wide32(t0, a.c0, b.c0);
+  mul_wide32(t1, a.c1, b.c1);
+  fp_add(sa, a.c0, a.c1);
+  fp_add(sb, b.c0, b.c1);
+  mul_wide32(m, sa, sb);
+  wide_sub_pp2(cw, t0, t1);
+  redc32(r.c0, cw);
+  wide_add(s01, t0, t1);
+  wide_sub_pp2(cw, m, s01);
+  redc32(r.c1, cw);
+}
+
+__global__ void chainG(const uint64_t *in, uint64_t *out, int iters) {
+  fp2 a, b;
+#pragma unroll
+  for (int i = 0; i < 6; i++) {
+    a.c0.v[i] = in[i] + threadIdx.x + blockIdx.x * 64; a.c1.v[i] = in[6+i];
+    b.c0.v[i] = in[6+i] ^ 0x5555; b.c1.v[i] = in[i] ^ 0x3333;
+  }
+  b.c0.v[5] %= BLS_P[5]; b.c1.v[5] %= BLS_P[5]; a.c1.v[5] %= BLS_P[5];
+  for (int q = 0; q < iters; q++) mul2_lazy(a, a, b);
+  uint64_t *o = out + 12 * (blockIdx.x * 64 + threadIdx.x);
+#pragma unroll
+  for (int i = 0; i < 6; i++) { o[i] = a.c0.v[i]; o[6+i] = a.c1.v[i]; }
+}
+
+// baseline for G: same fp2 chain with the production 12x32 CIOS
+__device__ __forceinline__ void mul2E(fp2 &r, const fp2 &a, const fp2 &b) {
+  fp t0, t1, s0, s1, m;
+  mulE(t0, a.c0, b.c0);
+  mulE(t1, a.c1, b.c1);
+  fp_add(s0, a.c0, a.c1);
+  fp_add(s1, b.c0, b.c1);
+  mulE(m, s0, s1);
+  fp_sub(m, m, t0);
+  fp_sub(m, m, t1);
+  fp_sub(r.c0, t0, t1);
+  r.c1 = m;
+}
+__global__ void chainG0(const uint64_t *in, uint64_t *out, int iters) {
+  fp2 a, b;
+#pragma unroll
+  for (int i = 0; i < 6; i++) {
+    a.c0.v[i] = in[i] + threadIdx.x + blockIdx.x * 64; a.c1.v[i] = in[6+i];
+    b.c0.v[i] = in[6+i] ^ 0x5555; b.c1.v[i] = in[i] ^ 0x3333;
+  }
+  b.c0.v[5] %= BLS_P[5]; b.c1.v[5] %= BLS_P[5]; a.c1.v[5] %= BLS_P[5];
+  for (int q = 0; q < iters; q++) mul2E(a, a, b);
+  uint64_t *o = out + 12 * (blockIdx.x * 64 + threadIdx.x);
+#pragma unroll
+  for (int i = 0; i < 6; i++) { o[i] = a.c0.v[i]; o[6+i] = a.c1.v[i]; }
+}
+
+// specialized 12x32 SOS squaring: 66 cross + 12 diag MACs (+ redc 144)
+// vs 144+144 for mulE(a,a)
+__device__ __forceinline__ void sqr_wide32(uint32_t t[25], const fp &a) {
+  const uint32_t *A = reinterpret_cast<const uint32_t *>(a.v);
+#pragma unroll
+  for (int i = 0; i < 25; i++) t[i] = 0;
+#pragma unroll
+  for (int i = 0; i < 11; i++) {
+    uint64_t c = 0;
+#pragma unroll
+    for (int j = i + 1; j < 12; j++) {
+      uint64_t s = (uint64_t)A[i] * A[j] + t[i + j] + (uint32_t)c;
+      t[i + j] = (uint32_t)s;
+      c = s >> 32;
+    }
+    t[i + 12] = (uint32_t)c;
+  }
+  uint32_t cc = 0;
+#pragma unroll
+  for (int i = 0; i < 24; i++) {
+    uint32_t nv = (t[i] << 1) | cc;
+    cc = t[i] >> 31;
+    t[i] = nv;
+  }
+  uint64_t c = 0;
+#pragma unroll
+  for (int i = 0; i < 12; i++) {
+    uint64_t s = (uint64_t)A[i] * A[i] + t[2 * i] + (uint32_t)c;
+    t[2 * i] = (uint32_t)s;
+    uint64_t s2 = (uint64_t)t[2 * i + 1] + (s >> 32);
+    t[2 * i + 1] = (uint32_t)s2;
+    c = s2 >> 32;
+  }
+}
+__device__ __forceinline__ void sqrH(fp &r, const fp &a) {
+  uint32_t t[25];
+  sqr_wide32(t, a);
+  redc32(r, t);
+}
+template <int USE_SQR>
+__global__ void chainH(const uint64_t *in, uint64_t *out, int iters) {
+  fp a;
+#pragma unroll
+  for (int i = 0; i < 6; i++) a.v[i] = in[i] + threadIdx.x + blockIdx.x * 64;
+  for (int q = 0; q < iters; q++) {
+    if (USE_SQR) sqrH(a, a);
+    else mulE(a, a, a);
+  }
+  uint64_t *o = out + 6 * (blockIdx.x * 64 + threadIdx.x);
+#pragma unroll
+  for (int i = 0; i < 6; i++) o[i] = a.v[i];
+}
+
 template <int V>
 __global__ void chain(const uint64_t *in, uint64_t *out, int iters) {
   fp a, b;
@@ -413,6 +645,43 @@ int main() {
     bool ok = true;
     for (int i = 0; i < 6; i++) ok &= (rd[i] == rs[i]);
     printf("D(first chain) == A: %s\n", ok ? "YES" : "NO");
+  }
+  auto bench = [&](const char *name, void (*kern)(const uint64_t *, uint64_t *, int), double per) {
+    hipEvent_t e0, e1; hipEventCreate(&e0); hipEventCreate(&e1);
+    for (int rep = 0; rep < 3; rep++) {
+      hipEventRecord(e0);
+      hipLaunchKernelGGL(kern, dim3(1024), dim3(64), 0, 0, d_in, d_out, ITERS);
+      hipEventRecord(e1);
+      hipError_t err = hipEventSynchronize(e1);
+      float ms; hipEventElapsedTime(&ms, e0, e1);
+      printf("%s rep %d: %.3f ms (%s) -> %.0f cyc/op, %.0f per fp_mul-equiv\n",
+             name, rep, ms, hipGetErrorString(err), ms * 1e-3 * 2.4e9 / ITERS,
+             ms * 1e-3 * 2.4e9 / ITERS / per);
+    }
+  };
+  bench("variant G0 (fp2 via 12x32 CIOS)", chainG0, 3.0);
+  bench("variant G  (fp2 LAZY 3 wide + 2 redc)", chainG, 3.0);
+  {
+    uint64_t rg[12], rg0[12];
+    hipLaunchKernelGGL(chainG0, dim3(1), dim3(64), 0, 0, d_in, d_out, 333);
+    hipMemcpy(rg0, d_out, 96, hipMemcpyDeviceToHost);
+    hipLaunchKernelGGL(chainG, dim3(1), dim3(64), 0, 0, d_in, d_out, 333);
+    hipMemcpy(rg, d_out, 96, hipMemcpyDeviceToHost);
+    bool ok = true;
+    for (int i = 0; i < 12; i++) ok &= (rg[i] == rg0[i]);
+    printf("G == G0 after 333 chained fp2 muls: %s\n", ok ? "YES" : "NO");
+  }
+  bench("variant H0 (square via mulE(a,a))", chainH<0>, 1.0);
+  bench("variant H  (specialized SOS sqr)", chainH<1>, 1.0);
+  {
+    uint64_t rh[6], rh0[6];
+    hipLaunchKernelGGL(chainH<0>, dim3(1), dim3(64), 0, 0, d_in, d_out, 444);
+    hipMemcpy(rh0, d_out, 48, hipMemcpyDeviceToHost);
+    hipLaunchKernelGGL(chainH<1>, dim3(1), dim3(64), 0, 0, d_in, d_out, 444);
+    hipMemcpy(rh, d_out, 48, hipMemcpyDeviceToHost);
+    bool ok = true;
+    for (int i = 0; i < 6; i++) ok &= (rh[i] == rh0[i]);
+    printf("H == H0 after 444 chained squarings: %s\n", ok ? "YES" : "NO");
   }
   // cross-check results equal
   uint64_t ra[6], rb[6];
