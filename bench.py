@@ -411,6 +411,53 @@ def main():
         torch.cuda.synchronize()
         shuffle_ms = (time.time() - tb) * 1e3
 
+    # C5 shape at N=1: 1,048,576 all-k=1 sets on ONE GPU (BASELINE
+    # configs[4]'s per-GPU workload; the driver's SCALE run shards the
+    # same shape over N GPUs). Reuses the 4096-key pool: signatures are
+    # the pool signature for msg-index mod pool — generation cost stays
+    # bounded while every set still carries a REAL valid signature.
+    c5_sets_per_sec = None
+    if rank == 0 and world == 1 and not DRY:
+        import numpy as _np
+
+        t0 = time.time()
+        n1m = 1 << 20
+        pool_msgs = bls_w["msgs"][: 32 * POOL]
+        pool_sigs = bls_w["sigs"][: 96 * POOL]
+        pool_pks = bls_w["pks"][: 96 * POOL]  # first POOL sets are k=1
+        reps = n1m // POOL
+        d5 = {
+            "msgs": ctx.upload(pool_msgs * reps),
+            "sigs": ctx.upload(pool_sigs * reps),
+            "pks": ctx.upload(pool_pks * reps),
+            "offsets": ctx.upload(
+                _np.arange(n1m + 1, dtype=_np.uint32).tobytes()
+            ),
+            "rands": ctx.upload(
+                _np.asarray(
+                    [((i * 0x9E3779B97F4A7C15 + 0xC0FFEE) | 1) & (2**64 - 1)
+                     for i in range(n1m)],
+                    dtype=_np.uint64,
+                ).tobytes()
+            ),
+        }
+        log(f"c5 1M-set workload staged in {time.time()-t0:.1f}s")
+        v = lib.m3x_bls_verify_sets_dev(
+            ctx.handle, d5["msgs"], d5["sigs"], d5["pks"], d5["offsets"],
+            d5["rands"], n1m)
+        assert v == 1
+        torch.cuda.synchronize()
+        tb = time.time()
+        v = lib.m3x_bls_verify_sets_dev(
+            ctx.handle, d5["msgs"], d5["sigs"], d5["pks"], d5["offsets"],
+            d5["rands"], n1m)
+        torch.cuda.synchronize()
+        c5_t = time.time() - tb
+        assert v == 1
+        c5_sets_per_sec = n1m / c5_t
+        for p in d5.values():
+            ctx.free(p)
+
     # split timing: one more pass of each, timed separately (for extras)
     sync()
     tb = time.time()
@@ -552,6 +599,7 @@ def main():
                 "c4_block_import_ms": c4_ms,
                 "registry_incremental_update_2048_ms": incr_ms,
                 "shuffle_1m_90rounds_ms": shuffle_ms,
+                "c5_1m_sets_1gpu_sets_per_sec": c5_sets_per_sec,
                 "cpu_sha_baseline": cpu_sha,
             },
             "roofline": roofline,

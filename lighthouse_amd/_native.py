@@ -306,6 +306,7 @@ class Ctx:
         "bls_miller",
         "bls_reduce",
         "bls_finish",
+        "bls_agg",
     ]
 
     def kernel_times(self):

@@ -246,6 +246,25 @@ __global__ __launch_bounds__(64, 1) void k_bls_miller(uint64_t n, BlsWork w) {
   w.fparts[i] = f;
 }
 
+// WAVE-SPLIT variant (round 2): 2n lanes — lane i = set i's high bit-half,
+// lane n+i = set i's low half. Doubles the wave count (2/SIMD at the C2
+// 64k-set shape) and halves each lane's dependent chain; the 2n partial
+// products feed the same GT reduction (their product = the n full
+// Millers' product). Each wave is role-uniform: no intra-wave divergence.
+__global__ __launch_bounds__(64, 1) void k_bls_miller_split(uint64_t n,
+                                                            BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 2 * n) return;
+  int role = lane < n ? 0 : 1;
+  uint64_t i = role == 0 ? lane : lane - n;
+  fp12m f;
+  if (*w.fail == 0)
+    miller_half(f, w.p_scaled[i], w.h2c[i], role);
+  else
+    f12_one(f);
+  w.fparts[lane] = f;
+}
+
 // two-stage GT-product reduction: each block folds its contiguous span of
 // per-set miller values (thread-strided local products, then an LDS tree)
 // into one output element; a second 1-block launch folds the partials.
@@ -420,7 +439,7 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t off_p = bytes; bytes += align(n * sizeof(g1j));
   uint64_t off_h = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
-  uint64_t off_f = bytes; bytes += align(n * sizeof(fp12m));
+  uint64_t off_f = bytes; bytes += align(2 * n * sizeof(fp12m)); // split miller: 2n partials
   uint64_t off_fail = bytes; bytes += 256;
   uint64_t off_sig_stage = bytes; bytes += align(256 * sizeof(g2j));
   uint64_t off_sum = bytes; bytes += align(sizeof(g2j));
@@ -447,12 +466,14 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   M3X_HIP_CHECK(hipMemsetAsync(w.fail, 0, 4, ctx->stream));
   M3X_HIP_CHECK(hipMemsetAsync(w.agg_count, 0, 4, ctx->stream));
   uint32_t blocks = (uint32_t)((n + 63) / 64);
-  m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
+  m3x::time_begin(ctx, M3X_K_BLS_AGG);
   hipLaunchKernelGGL(k_bls_scan_agg, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint32_t *)offs_dev, n, w);
   hipLaunchKernelGGL(k_bls_aggregate_w, dim3((uint32_t)n), dim3(64), 0,
                      ctx->stream, (const uint8_t *)pks_dev,
                      (const uint32_t *)offs_dev, w);
+  m3x::time_end(ctx, M3X_K_BLS_AGG);
+  m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
   hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
                      (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
@@ -472,19 +493,23 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   // small batches are latency-bound on the per-lane kernel: go wave-per-set
   uint64_t small_thresh = 2048; // measured crossover (tmp_bench/c4probe)
   if (const char *e = getenv("M3X_SMALL_MILLER")) small_thresh = strtoull(e, nullptr, 10);
-  if (n <= small_thresh)
+  uint64_t n_parts = n; // fp12 partials feeding the GT reduce
+  if (n <= small_thresh) {
     hipLaunchKernelGGL(k_bls_miller_small, dim3((uint32_t)n), dim3(64), 0,
                        ctx->stream, n, w);
-  else
-    hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream,
-                       n, w);
+  } else {
+    n_parts = 2 * n; // wave-split: two half-Millers per set
+    uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
+    hipLaunchKernelGGL(k_bls_miller_split, dim3(blocks2), dim3(64), 0,
+                       ctx->stream, n, w);
+  }
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
   DBG_STEP(ctx, "miller");
-  uint32_t rblocks = (uint32_t)((n + 255) / 256);
+  uint32_t rblocks = (uint32_t)((n_parts + 255) / 256);
   if (rblocks > 256) rblocks = 256;
   m3x::time_begin(ctx, M3X_K_BLS_REDUCE);
   hipLaunchKernelGGL(k_bls_reduce_gt, dim3(rblocks), dim3(256), 0,
-                     ctx->stream, w.fparts, n, w.gt_stage);
+                     ctx->stream, w.fparts, n_parts, w.gt_stage);
   hipLaunchKernelGGL(k_bls_reduce_gt, dim3(1), dim3(256), 0, ctx->stream,
                      w.gt_stage, (uint64_t)rblocks, w.gt_parts);
   hipLaunchKernelGGL(k_bls_reduce_sig, dim3(rblocks), dim3(256), 0,

@@ -14,7 +14,8 @@ enum m3x_kernel_id {
   M3X_K_BLS_MILLER = 5,
   M3X_K_BLS_REDUCE = 6,
   M3X_K_BLS_FINISH = 7,
-  M3X_K_COUNT = 8
+  M3X_K_BLS_AGG = 8,
+  M3X_K_COUNT = 9
 };
 
 struct m3x_ctx {
