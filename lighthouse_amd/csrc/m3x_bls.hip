@@ -35,7 +35,10 @@ struct BlsWork {
   uint32_t *agg_count;
   g1j *p_scaled;   // [n] r_i * aggregate pubkey (Jacobian)
   g2j *h2c;        // [n] hash_to_curve(msg), Jacobian (no inversion)
+  uint8_t *uni;    // [n*256] expand_message_xmd output (h2c pass 1)
+  g2j *h2c_pts;    // [2n] per-point sswu+iso outputs (h2c pass 2)
   g2j *rsig;       // [n] r_i * sigma (jacobian)
+  g2j *sig_aff;    // [n] decompressed sigma (z=1 affine; z=0 infinity)
   fp12m *fparts;   // [n] per-set miller values
   int *fail;       // [1]
   g2j *sig_stage;  // [256] stage-1 partial sums
@@ -99,8 +102,11 @@ __global__ __launch_bounds__(64, 1) void k_bls_aggregate_w(
     const uint8_t *__restrict__ pks, const uint32_t *__restrict__ offs,
     BlsWork w) {
   __shared__ g1j lds[64];
-  if (blockIdx.x >= *w.agg_count) return;
-  uint64_t set = w.agg_idx[blockIdx.x];
+  // grid-stride over the aggregate list: the grid is FIXED (8192 blocks)
+  // so a k=1-only workload costs ~nothing — launching n blocks of
+  // early-exits measured 183 ms at n=1M (rocprof r02)
+  for (uint32_t b = blockIdx.x; b < *w.agg_count; b += gridDim.x) {
+  uint64_t set = w.agg_idx[b];
   uint32_t k0 = offs[set], k1 = offs[set + 1];
   int lane = threadIdx.x;
   g1j acc;
@@ -128,60 +134,132 @@ __global__ __launch_bounds__(64, 1) void k_bls_aggregate_w(
     __syncthreads();
   }
   if (lane == 0) w.apk[set] = lds[0];
+  __syncthreads(); // next grid-stride iteration reuses lds
+  }
 }
 
-// per-set prepare: sig decompress + subgroup check, pk aggregation,
-// r-scaling of both sides
-__global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict__ sigs,
-                              const uint8_t *__restrict__ pks,
-                              const uint32_t *__restrict__ offs,
-                              const uint64_t *__restrict__ rands, uint64_t n,
-                              BlsWork w) {
+// ---- WAVE-SPLIT prepare (round 2; was one ~30ms/lane serial kernel) ----
+// pass 1: decompress sigma (n lanes; the sqrt pow chain dominates)
+__global__ __launch_bounds__(64, 1) void k_bls_sigdec(
+    const uint8_t *__restrict__ sigs, uint64_t n, BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   g2a sig;
   if (g2_decompress(sig, sigs + 96 * i) != 0) {
     atomicOr(w.fail, 1);
+    sig.inf = 1; // harmless placeholder; verdict is already forced false
+  }
+  g2j out;
+  if (sig.inf) {
+    fp2_zero(out.x);
+    fp2_zero(out.y);
+    fp2_zero(out.z);
+  } else {
+    out.x = sig.x;
+    out.y = sig.y;
+    fp2_one(out.z);
+  }
+  w.sig_aff[i] = out;
+}
+
+// [k]B for an affine G2 base with MIXED adds (24 vs 43 fp-muls per add);
+// MSB-first double-and-add. Separate per-scalar chains with mixed adds
+// cost LESS total than the round-1 shared-doubling chain of FULL adds.
+__device__ inline void g2_mul_u64_aff(g2j &r, const g2a &base, uint64_t k) {
+  g2j acc;
+  fp2_zero(acc.x);
+  fp2_zero(acc.y);
+  fp2_zero(acc.z);
+  for (int b = 63; b >= 0; b--) {
+    g2j_dbl(acc, acc);
+    if ((k >> b) & 1) g2j_add_aff(acc, acc, base);
+  }
+  r = acc;
+}
+
+// [k]B for an affine G1 base with mixed adds
+__device__ inline void g1_mul_u64_aff(g1j &r, const g1a &base, uint64_t k) {
+  g1j acc;
+  fp_zero(acc.x);
+  fp_zero(acc.y);
+  fp_zero(acc.z);
+  for (int b = 63; b >= 0; b--) {
+    g1j_dbl(acc, acc);
+    if ((k >> b) & 1) g1j_add_aff(acc, acc, base);
+  }
+  r = acc;
+}
+
+// pass 2: the scalar-mult work, TWO wave-uniform classes over 2n lanes
+// (2 waves/SIMD at the C2 shape):
+//   class A (lane i):   rsig[i] = [r_i] sigma_i
+//   class B (lane n+i): p_scaled[i] = [r_i] apk_i  AND the deferred
+//                       psi-subgroup check psi(sigma) == -[|x|]sigma
+//                       (blst.rs:73-77)
+__global__ __launch_bounds__(64, 1) void k_bls_prep_mults(
+    const uint8_t *__restrict__ pks, const uint32_t *__restrict__ offs,
+    const uint64_t *__restrict__ rands, uint64_t n, BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 2 * n) return;
+  if (lane < n) {
+    uint64_t i = lane;
+    g2j sj = w.sig_aff[i];
+    if (g2j_is_inf(sj)) {
+      // infinity contributes nothing to the signature sum
+      fp2_zero(w.rsig[i].x);
+      fp2_zero(w.rsig[i].y);
+      fp2_zero(w.rsig[i].z);
+      return;
+    }
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j rs;
+    g2_mul_u64_aff(rs, sig, rands[i]);
+    w.rsig[i] = rs;
     return;
   }
+  uint64_t i = lane - n;
   uint32_t k0 = offs[i], k1 = offs[i + 1];
   if (k1 <= k0) {
     atomicOr(w.fail, 1);
     return;
   }
-  g1j apk;
+  g1j rp;
   if (k1 - k0 == 1) {
     g1a pk;
     if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k0) != 0) {
       atomicOr(w.fail, 1);
       return;
     }
-    g1j_from_aff(apk, pk);
+    if (pk.inf) { // aggregate at infinity -> invalid
+      atomicOr(w.fail, 1);
+      return;
+    }
+    g1_mul_u64_aff(rp, pk, rands[i]); // mixed adds on the affine base
   } else {
-    apk = w.apk[i]; // precomputed by k_bls_aggregate_w
-  }
-  if (g1j_is_inf(apk)) { // aggregate at infinity -> invalid
-    atomicOr(w.fail, 1);
-    return;
-  }
-  uint8_t rbe[8];
+    g1j apk = w.apk[i]; // precomputed by k_bls_aggregate_w
+    if (g1j_is_inf(apk)) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    uint8_t rbe[8];
 #pragma unroll
-  for (int b = 0; b < 8; b++) rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
-  g1j rp;
-  g1j_mul_be_j(rp, apk, rbe, 8); // P stays Jacobian end-to-end
+    for (int b = 0; b < 8; b++)
+      rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
+    g1j_mul_be_j(rp, apk, rbe, 8);
+  }
   w.p_scaled[i] = rp;
-  if (sig.inf) {
-    // infinity is a valid subgroup element; contributes nothing
-    fp2_zero(w.rsig[i].x);
-    fp2_zero(w.rsig[i].y);
-    fp2_zero(w.rsig[i].z);
-  } else {
-    // [r]sigma and the psi subgroup check's [|x|]sigma share sigma's
-    // doubling chain (blst.rs:73-77 deferred subgroup check)
-    g2j rsig_j, xsig_j;
-    g2j_mul2_u64(rsig_j, xsig_j, sig, rands[i], BLS_X_ABS);
-    w.rsig[i] = rsig_j;
-    // psi(sigma) must equal -[|x|]sigma (x < 0): cross-multiplied compare
+  // deferred subgroup check (skipped for infinity: valid element)
+  g2j sj = w.sig_aff[i];
+  if (!g2j_is_inf(sj)) {
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j xsig_j;
+    g2_mul_u64_aff(xsig_j, sig, BLS_X_ABS);
     g2a ps;
     psi_g2(ps, sig);
     bool ok;
@@ -196,18 +274,56 @@ __global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict
       fp2_mul(ly, ps.y, z3);
       ok = fp2_eq(lx, xsig_j.x) && fp2_eq(ly, ny);
     }
-    if (!ok) {
-      atomicOr(w.fail, 1);
-      return;
-    }
+    if (!ok) atomicOr(w.fail, 1);
   }
 }
 
-__global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
+__global__ __launch_bounds__(64, 1) void k_bls_h2c(__global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
                           BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   m3xb::h2c_g2(w.h2c[i], msgs + 32 * i);
+}
+
+// ---- WAVE-SPLIT h2c (round 2): three passes so the heavy SSWU work runs
+// at 2n lanes (2 waves/SIMD at the C2 shape) and each lane's dependent
+// chain is one point, not two ----
+__global__ __launch_bounds__(64, 1) void k_bls_h2c_expand(
+    const uint8_t *__restrict__ msgs, uint64_t n, BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  expand_message_xmd32(msgs + 32 * i, w.uni + 256 * i);
+}
+
+__global__ __launch_bounds__(64, 2) void k_bls_h2c_map(uint64_t n,
+                                                       BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 2 * n) return;
+  int role = lane < n ? 0 : 1; // which of the two RO points
+  uint64_t i = role == 0 ? lane : lane - n;
+  const uint8_t *uni = w.uni + 256 * i + 128 * role;
+  fp2 u;
+  h2f_from_be64(u.c0, uni);
+  h2f_from_be64(u.c1, uni + 64);
+  g2a q;
+  sswu_g2(q, u);
+  g2j pt, cleared;
+  iso_map_g2_j(pt, q);
+  // cofactor clearing is a homomorphism (h_eff scalar mult + psi), so
+  // clearing each RO point separately at 2n lanes (2 waves/SIMD hiding
+  // the dependent dbl-chain latency) and summing afterwards equals
+  // clearing the sum — the fin pass is then one mixed add per set
+  clear_cofactor_g2j(cleared, pt);
+  w.h2c_pts[lane] = cleared;
+}
+
+__global__ __launch_bounds__(64, 1) void k_bls_h2c_fin(uint64_t n,
+                                                       BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  g2j s = w.h2c_pts[i];
+  g2j_add(s, s, w.h2c_pts[n + i]);
+  w.h2c[i] = s;
 }
 
 // small-batch variant: ONE WAVE PER SET (cooperative miller_w). At tiny n
@@ -251,8 +367,12 @@ __global__ __launch_bounds__(64, 1) void k_bls_miller(uint64_t n, BlsWork w) {
 // 64k-set shape) and halves each lane's dependent chain; the 2n partial
 // products feed the same GT reduction (their product = the n full
 // Millers' product). Each wave is role-uniform: no intra-wave divergence.
-__global__ __launch_bounds__(64, 1) void k_bls_miller_split(uint64_t n,
-                                                            BlsWork w) {
+__global__ __launch_bounds__(64, 2)
+__attribute__((amdgpu_waves_per_eu(2))) void k_bls_miller_split(
+    uint64_t n, BlsWork w) {
+  // launch_bounds min-waves 2: cap the allocation at 256 VGPRs so the two
+  // half-Miller waves of a SIMD actually co-reside (the whole point of
+  // the split; at the default the allocator takes 511 -> 1 wave/SIMD)
   uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (lane >= 2 * n) return;
   int role = lane < n ? 0 : 1;
@@ -438,7 +558,10 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t off_aggcnt = bytes; bytes += 256;
   uint64_t off_p = bytes; bytes += align(n * sizeof(g1j));
   uint64_t off_h = bytes; bytes += align(n * sizeof(g2j));
+  uint64_t off_uni = bytes; bytes += align(n * 256);
+  uint64_t off_hpts = bytes; bytes += align(2 * n * sizeof(g2j));
   uint64_t off_r = bytes; bytes += align(n * sizeof(g2j));
+  uint64_t off_sa = bytes; bytes += align(n * sizeof(g2j));
   uint64_t off_f = bytes; bytes += align(2 * n * sizeof(fp12m)); // split miller: 2n partials
   uint64_t off_fail = bytes; bytes += 256;
   uint64_t off_sig_stage = bytes; bytes += align(256 * sizeof(g2j));
@@ -455,7 +578,10 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   w.agg_count = reinterpret_cast<uint32_t *>(base + off_aggcnt);
   w.p_scaled = reinterpret_cast<g1j *>(base + off_p);
   w.h2c = reinterpret_cast<g2j *>(base + off_h);
+  w.uni = base + off_uni;
+  w.h2c_pts = reinterpret_cast<g2j *>(base + off_hpts);
   w.rsig = reinterpret_cast<g2j *>(base + off_r);
+  w.sig_aff = reinterpret_cast<g2j *>(base + off_sa);
   w.fparts = reinterpret_cast<fp12m *>(base + off_f);
   w.fail = reinterpret_cast<int *>(base + off_fail);
   w.sig_stage = reinterpret_cast<g2j *>(base + off_sig_stage);
@@ -469,22 +595,40 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   m3x::time_begin(ctx, M3X_K_BLS_AGG);
   hipLaunchKernelGGL(k_bls_scan_agg, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint32_t *)offs_dev, n, w);
-  hipLaunchKernelGGL(k_bls_aggregate_w, dim3((uint32_t)n), dim3(64), 0,
-                     ctx->stream, (const uint8_t *)pks_dev,
-                     (const uint32_t *)offs_dev, w);
+  {
+    uint32_t agg_blocks = n < 8192 ? (uint32_t)n : 8192;
+    hipLaunchKernelGGL(k_bls_aggregate_w, dim3(agg_blocks), dim3(64), 0,
+                       ctx->stream, (const uint8_t *)pks_dev,
+                       (const uint32_t *)offs_dev, w);
+  }
   m3x::time_end(ctx, M3X_K_BLS_AGG);
   m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
-  hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0, ctx->stream,
-                     (const uint8_t *)sigs_dev, (const uint8_t *)pks_dev,
-                     (const uint32_t *)offs_dev, (const uint64_t *)rands_dev,
-                     n, w);
+  hipLaunchKernelGGL(k_bls_sigdec, dim3(blocks), dim3(64), 0, ctx->stream,
+                     (const uint8_t *)sigs_dev, n, w);
+  {
+    uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
+    hipLaunchKernelGGL(k_bls_prep_mults, dim3(blocks2), dim3(64), 0,
+                       ctx->stream, (const uint8_t *)pks_dev,
+                       (const uint32_t *)offs_dev,
+                       (const uint64_t *)rands_dev, n, w);
+  }
   m3x::time_end(ctx, M3X_K_BLS_PREPARE);
   DBG_STEP(ctx, "prepare");
   // h2c is independent of prepare: run it on the second stream so the two
   // ~1-wave/SIMD kernels co-reside (both fit at 2 waves/SIMD by VGPR count)
   m3x::time_begin_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
-  hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream2,
-                     (const uint8_t *)msgs_dev, n, w);
+  if (n > 2048) {
+    uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
+    hipLaunchKernelGGL(k_bls_h2c_expand, dim3(blocks), dim3(64), 0,
+                       ctx->stream2, (const uint8_t *)msgs_dev, n, w);
+    hipLaunchKernelGGL(k_bls_h2c_map, dim3(blocks2), dim3(64), 0,
+                       ctx->stream2, n, w);
+    hipLaunchKernelGGL(k_bls_h2c_fin, dim3(blocks), dim3(64), 0,
+                       ctx->stream2, n, w);
+  } else {
+    hipLaunchKernelGGL(k_bls_h2c, dim3(blocks), dim3(64), 0, ctx->stream2,
+                       (const uint8_t *)msgs_dev, n, w);
+  }
   m3x::time_end_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
   M3X_HIP_CHECK(hipEventRecord(ctx->ev_s2, ctx->stream2));
   M3X_HIP_CHECK(hipStreamWaitEvent(ctx->stream, ctx->ev_s2, 0));
@@ -494,14 +638,23 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   uint64_t small_thresh = 2048; // measured crossover (tmp_bench/c4probe)
   if (const char *e = getenv("M3X_SMALL_MILLER")) small_thresh = strtoull(e, nullptr, 10);
   uint64_t n_parts = n; // fp12 partials feeding the GT reduce
+  static int use_split = -1;
+  if (use_split < 0) {
+    const char *e = getenv("M3X_MILLER_SPLIT");
+    use_split = (e && e[0] == '1') ? 1 : 0; // measured SLOWER at C2 (57
+    // vs 44 ms: the 256-VGPR cap spills more than co-residency saves)
+  }
   if (n <= small_thresh) {
     hipLaunchKernelGGL(k_bls_miller_small, dim3((uint32_t)n), dim3(64), 0,
                        ctx->stream, n, w);
-  } else {
+  } else if (use_split) {
     n_parts = 2 * n; // wave-split: two half-Millers per set
     uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
     hipLaunchKernelGGL(k_bls_miller_split, dim3(blocks2), dim3(64), 0,
                        ctx->stream, n, w);
+  } else {
+    hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream,
+                       n, w);
   }
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
   DBG_STEP(ctx, "miller");
