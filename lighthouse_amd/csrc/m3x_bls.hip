@@ -278,7 +278,7 @@ __global__ __launch_bounds__(64, 1) void k_bls_prep_mults(
   }
 }
 
-__global__ __launch_bounds__(64, 1) void k_bls_h2c(__global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
+__global__ __launch_bounds__(64, 1) void k_bls_h2c(const uint8_t *__restrict__ msgs, uint64_t n,
                           BlsWork w) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
