@@ -1743,12 +1743,75 @@ __device__ inline void f12_sqr_w(fp12m &a, f12w_ws &ws, int lane) {
 }
 
 
-// r = a^|x|; r distinct from a
+// cyclotomic squaring (Granger-Scott), cooperative — valid ONLY for
+// elements of the cyclotomic subgroup (anything after the easy part,
+// i.e. every final-exp pow-chain input). 9 fp2 squarings across lanes
+// (vs 21 products in f12_sqr_w); the three Fp4 = Fp2[w^3] subalgebras
+// (g0,g3) (g1,g4) (g2,g5) are squared and recombined as
+//   h_even = 3*t0 - 2*g_even,  h_odd = 3*cross(*xi for h1) + 2*g_odd.
+// Formulas validated against the generic squaring on random cyclotomic
+// elements in Python (see round-2 session notes / DESIGN.md).
+__constant__ int F12_CYC_OUT[6] = {0, 3, 2, 5, 4, 1};
+__constant__ int F12_CYC_PAIR[6] = {0, 0, 1, 1, 2, 2};
+__constant__ int F12_CYC_KIND[6] = {0, 1, 0, 1, 0, 1}; // 0 = t0, 1 = cross
+
+__device__ inline void f12_sqr_cyc_w(fp12m &a, f12w_ws &ws, int lane) {
+  if (lane < 9) {
+    int pair = lane / 3, which = lane % 3;
+    fp2 a0, a1, v, sq;
+    f12_get(a, pair, a0);
+    f12_get(a, pair + 3, a1);
+    if (which == 0)
+      v = a0;
+    else if (which == 1)
+      v = a1;
+    else
+      fp2_add(v, a0, a1);
+    fp2_sqr(sq, v);
+    ws.t[lane] = sq;
+  }
+  f12w_sync();
+  if (lane < 6) {
+    int out = F12_CYC_OUT[lane];
+    int pr = F12_CYC_PAIR[lane];
+    fp2 s0 = ws.t[3 * pr], s1 = ws.t[3 * pr + 1], ss = ws.t[3 * pr + 2];
+    fp2 t, h, gcur;
+    f12_get(a, out, gcur);
+    if (F12_CYC_KIND[lane] == 0) {
+      fp2_mul_xi(t, s1);
+      fp2_add(t, t, s0); /* t0 = a0^2 + xi*a1^2 */
+      fp2 t3;
+      fp2_dbl(t3, t);
+      fp2_add(t3, t3, t); /* 3*t0 */
+      fp2_dbl(gcur, gcur);
+      fp2_sub(h, t3, gcur); /* 3*t0 - 2*g */
+    } else {
+      fp2_sub(t, ss, s0);
+      fp2_sub(t, t, s1); /* cross = 2*a0*a1 */
+      if (out == 1) fp2_mul_xi(t, t);
+      fp2 t3;
+      fp2_dbl(t3, t);
+      fp2_add(t3, t3, t); /* 3*cross */
+      fp2_dbl(gcur, gcur);
+      fp2_add(h, t3, gcur); /* 3*cross + 2*g */
+    }
+    ws.t[18 + lane] = h; /* stage outputs (a still being read) */
+  }
+  f12w_sync();
+  if (lane < 6) {
+    int out = F12_CYC_OUT[lane];
+    f12_set(a, out, ws.t[18 + lane]);
+  }
+  f12w_sync();
+}
+
+// r = a^|x|; r distinct from a. a and all intermediates are cyclotomic
+// here (final-exp chains), so the Granger-Scott squaring applies.
 __device__ inline void f12_pow_xabs_w(fp12m &r, const fp12m &a, f12w_ws &ws,
                                       int lane) {
   f12_copy_w(r, a, lane);
   for (int b = 62; b >= 0; b--) {
-    f12_sqr_w(r, ws, lane);
+    f12_sqr_cyc_w(r, ws, lane);
     if ((BLS_X_ABS >> b) & 1) f12_mul_w(r, r, a, ws, lane);
   }
 }
