@@ -352,29 +352,17 @@ __global__ __launch_bounds__(64, 1) void k_bls_miller(uint64_t n, BlsWork w) {
   // fp12 state stays in thread-local scratch: an LDS-resident variant
   // measured 2x SLOWER (123ms vs 63ms on C2) — the L1/L2-cached spill
   // traffic beats per-limb ds_read latency for this access pattern.
-  //
-  // Round 2: the wave's 64 per-set Miller values fold to ONE partial
-  // product via an LDS tree before leaving the kernel — the GT reduce
-  // then reads 1 fp12 per wave instead of 64 (36MB less HBM traffic and
-  // a 64x smaller stage-1 reduce). Lanes past n contribute the identity.
-  __shared__ fp12m lds[64];
+  // (An in-wave LDS tree fold of the 64 per-set values also regressed:
+  // +5ms in the kernel and the 64x-smaller reduce became LATENCY-bound
+  // at 4 blocks — reverted; measured round 2.)
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int lane = threadIdx.x;
+  if (i >= n) return;
   fp12m f, tmp;
-  if (i < n && *w.fail == 0)
+  if (*w.fail == 0)
     miller_raw(f, tmp, w.p_scaled[i], w.h2c[i]);
   else
     f12_one(f);
-  f12_copy(lds[lane], f);
-  __syncthreads();
-  for (int s = 32; s > 0; s >>= 1) {
-    if (lane < s) {
-      f12_mul_nn(tmp, lds[lane], lds[lane + s]);
-      f12_copy(lds[lane], tmp);
-    }
-    __syncthreads();
-  }
-  if (lane == 0) w.fparts[blockIdx.x] = lds[0];
+  w.fparts[i] = f;
 }
 
 // WAVE-SPLIT variant (round 2): 2n lanes — lane i = set i's high bit-half,
@@ -670,7 +658,6 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   } else {
     hipLaunchKernelGGL(k_bls_miller, dim3(blocks), dim3(64), 0, ctx->stream,
                        n, w);
-    n_parts = blocks; // one in-wave-folded partial per block
   }
   m3x::time_end(ctx, M3X_K_BLS_MILLER);
   DBG_STEP(ctx, "miller");
