@@ -138,6 +138,83 @@ __global__ __launch_bounds__(64, 1) void k_bls_aggregate_w(
   }
 }
 
+// FUSED per-set prepare (kept for LARGE batches): at high occupancy the
+// kernel is ISSUE-bound and the shared-doubling dual-scalar chain does
+// less total work than the split form (rocprof r02g: 450 vs 657+116 ms
+// at 1M sets); the split form wins in the small-batch LATENCY regime
+// (32 vs 37 ms at 64k). Dispatch picks by n.
+__global__ __launch_bounds__(64, 1) void k_bls_prepare(const uint8_t *__restrict__ sigs,
+                              const uint8_t *__restrict__ pks,
+                              const uint32_t *__restrict__ offs,
+                              const uint64_t *__restrict__ rands, uint64_t n,
+                              BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  g2a sig;
+  if (g2_decompress(sig, sigs + 96 * i) != 0) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  uint32_t k0 = offs[i], k1 = offs[i + 1];
+  if (k1 <= k0) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  g1j apk;
+  if (k1 - k0 == 1) {
+    g1a pk;
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k0) != 0) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    g1j_from_aff(apk, pk);
+  } else {
+    apk = w.apk[i]; // precomputed by k_bls_aggregate_w
+  }
+  if (g1j_is_inf(apk)) { // aggregate at infinity -> invalid
+    atomicOr(w.fail, 1);
+    return;
+  }
+  uint8_t rbe[8];
+#pragma unroll
+  for (int b = 0; b < 8; b++) rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
+  g1j rp;
+  g1j_mul_be_j(rp, apk, rbe, 8); // P stays Jacobian end-to-end
+  w.p_scaled[i] = rp;
+  if (sig.inf) {
+    // infinity is a valid subgroup element; contributes nothing
+    fp2_zero(w.rsig[i].x);
+    fp2_zero(w.rsig[i].y);
+    fp2_zero(w.rsig[i].z);
+  } else {
+    // [r]sigma and the psi subgroup check's [|x|]sigma share sigma's
+    // doubling chain (blst.rs:73-77 deferred subgroup check)
+    g2j rsig_j, xsig_j;
+    g2j_mul2_u64(rsig_j, xsig_j, sig, rands[i], BLS_X_ABS);
+    w.rsig[i] = rsig_j;
+    // psi(sigma) must equal -[|x|]sigma (x < 0): cross-multiplied compare
+    g2a ps;
+    psi_g2(ps, sig);
+    bool ok;
+    if (g2j_is_inf(xsig_j)) {
+      ok = false;
+    } else {
+      fp2 z2, z3, lx, ly, ny;
+      fp2_sqr(z2, xsig_j.z);
+      fp2_mul(z3, z2, xsig_j.z);
+      fp2_mul(lx, ps.x, z2);
+      fp2_neg(ny, xsig_j.y);
+      fp2_mul(ly, ps.y, z3);
+      ok = fp2_eq(lx, xsig_j.x) && fp2_eq(ly, ny);
+    }
+    if (!ok) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+  }
+}
+
+
 // ---- WAVE-SPLIT prepare (round 2; was one ~30ms/lane serial kernel) ----
 // pass 1: decompress sigma (n lanes; the sqrt pow chain dominates)
 __global__ __launch_bounds__(64, 1) void k_bls_sigdec(
@@ -599,16 +676,23 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   hipLaunchKernelGGL(k_bls_scan_agg, dim3(blocks), dim3(64), 0, ctx->stream,
                      (const uint32_t *)offs_dev, n, w);
   {
-    uint32_t agg_blocks = n < 8192 ? (uint32_t)n : 8192;
+    uint32_t agg_blocks = n < 32768 ? (uint32_t)n : 32768;
     hipLaunchKernelGGL(k_bls_aggregate_w, dim3(agg_blocks), dim3(64), 0,
                        ctx->stream, (const uint8_t *)pks_dev,
                        (const uint32_t *)offs_dev, w);
   }
   m3x::time_end(ctx, M3X_K_BLS_AGG);
   m3x::time_begin(ctx, M3X_K_BLS_PREPARE);
-  hipLaunchKernelGGL(k_bls_sigdec, dim3(blocks), dim3(64), 0, ctx->stream,
-                     (const uint8_t *)sigs_dev, n, w);
-  {
+  if (n > (1ull << 18)) {
+    // issue-bound regime: the fused shared-chain kernel does less work
+    hipLaunchKernelGGL(k_bls_prepare, dim3(blocks), dim3(64), 0,
+                       ctx->stream, (const uint8_t *)sigs_dev,
+                       (const uint8_t *)pks_dev, (const uint32_t *)offs_dev,
+                       (const uint64_t *)rands_dev, n, w);
+  } else {
+    // latency regime: decompress pass + two wave-uniform mult classes
+    hipLaunchKernelGGL(k_bls_sigdec, dim3(blocks), dim3(64), 0, ctx->stream,
+                       (const uint8_t *)sigs_dev, n, w);
     uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
     hipLaunchKernelGGL(k_bls_prep_mults, dim3(blocks2), dim3(64), 0,
                        ctx->stream, (const uint8_t *)pks_dev,
