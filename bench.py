@@ -357,42 +357,25 @@ def main():
                 .tobytes()
             ),
         }
-        # The reference pipelines signature verification and state-root
-        # computation (block_verification.rs: payload verification is
-        # already concurrent with per_block_processing) — mirror that:
-        # BLS on the main context, the state root on a second context's
-        # streams, joined at the end.
-        from concurrent.futures import ThreadPoolExecutor
-
-        ctx2 = _native.Ctx(local_rank)
-
-        def c4_bls():
-            return lib.m3x_bls_verify_sets_dev(
-                ctx.handle, d4["msgs"], d4["sigs"], d4["pks"],
-                d4["offsets"], d4["rands"], len(idx4))
-
-        def c4_root():
-            rc = lib.m3x_validator_subtree_root_dev(
-                ctx2.handle, ssz_dev, per, sub_depth, out32)
-            assert rc == 0, rc
-            reg = ctx2.finalize_root(out32.raw, sub_depth, 40, N_VALIDATORS)
-            return bs.state_root(st, ctx=ctx2, dev=devmap, registry_root=reg)
-
-        with ThreadPoolExecutor(2) as pool:
-            # warm
-            v = pool.submit(c4_bls).result()
-            pool.submit(c4_root).result()
-            torch.cuda.synchronize()
-            tb = time.time()
-            f1 = pool.submit(c4_bls)
-            f2 = pool.submit(c4_root)
-            v = f1.result()
-            root4 = f2.result()
-            torch.cuda.synchronize()
-            c4_ms = (time.time() - tb) * 1e3
+        # (A two-context overlapped variant — BLS on one context, state
+        # root on another, mirroring the reference's concurrent block
+        # pipeline — measured NO win and extra variance on this path:
+        # the root's many small launches interleave anyway. Sequential
+        # kept for a clean number.)
+        for _ in range(1):
+            v = lib.m3x_bls_verify_sets_dev(
+                ctx.handle, d4["msgs"], d4["sigs"], d4["pks"], d4["offsets"],
+                d4["rands"], len(idx4))
+            merkle_step()
+        torch.cuda.synchronize()
+        tb = time.time()
+        v = lib.m3x_bls_verify_sets_dev(
+            ctx.handle, d4["msgs"], d4["sigs"], d4["pks"], d4["offsets"],
+            d4["rands"], len(idx4))
+        merkle_step()
+        torch.cuda.synchronize()
+        c4_ms = (time.time() - tb) * 1e3
         assert v == 1
-        assert root4 == full_root, "c4 overlapped state root mismatch"
-        ctx2.close()
 
     # incremental registry cache: per-block delta rehash
     incr_ms = None
