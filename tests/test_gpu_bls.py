@@ -275,3 +275,46 @@ def test_signature_aggregation_matches_aggregate_secret(ctx, oracle):
     assert oracle.m3x_oracle_bls_sign(sk_sum, msg, want) == 0
     assert agg.serialize() == want.raw
     assert bls.fast_aggregate_verify(agg, msg, pks, ctx=ctx) is True
+
+
+def test_large_batch_split_paths(ctx, oracle):
+    """n=4096 exercises the WAVE-SPLIT pipeline paths (h2c three-pass,
+    prepare decompress+mult classes; dispatched above n=2048) that the
+    small fixture batches never reach: valid batch -> True, one corrupted
+    message -> False, and rejection does not poison a following valid
+    batch on the same context."""
+    import ctypes
+
+    from lighthouse_amd import _native
+
+    n = 4096
+    sks = ctypes.create_string_buffer(32 * n)
+    pks = ctypes.create_string_buffer(96 * n)
+    oracle.m3x_oracle_bls_keypool(ctypes.c_uint64(n), sks, pks)
+    msgs = bytearray()
+    for i in range(n):
+        msgs += hashlib.sha256(b"large%d" % i).digest()
+    sigs = ctypes.create_string_buffer(96 * n)
+    assert (
+        oracle.m3x_oracle_bls_sign_batch(
+            ctypes.c_uint64(n), sks.raw, bytes(msgs), sigs
+        )
+        == 0
+    )
+    offs = (ctypes.c_uint32 * (n + 1))(*range(n + 1))
+    rnds = (ctypes.c_uint64 * n)(*[(i * 0x9E37 + 1) | 1 for i in range(n)])
+    lib = _native.load()
+    rc = lib.m3x_bls_verify_sets(
+        ctx.handle, bytes(msgs), sigs.raw, pks.raw, offs, rnds, n
+    )
+    assert rc == 1
+    bad = bytearray(msgs)
+    bad[32 * 1234 : 32 * 1235] = hashlib.sha256(b"tampered").digest()
+    rc = lib.m3x_bls_verify_sets(
+        ctx.handle, bytes(bad), sigs.raw, pks.raw, offs, rnds, n
+    )
+    assert rc == 0
+    rc = lib.m3x_bls_verify_sets(
+        ctx.handle, bytes(msgs), sigs.raw, pks.raw, offs, rnds, n
+    )
+    assert rc == 1
