@@ -627,6 +627,135 @@ __global__ void k_bls_expand_dst(const uint8_t *__restrict__ msg,
   expand_message_xmd_gen(msg, msg_len, dst, dst_len, len_in_bytes, out);
 }
 
+
+// LATENCY-REGIME register-budget twins: at n <= 2^18 these kernels run
+// 1-2 waves/SIMD, so granting each wave the idle register file (512 or
+// 256 VGPRs vs the ~130 the default allocation picks) trades nothing
+// and removes scratch spill round-trips from the serial chains. The
+// default-budget forms stay for the high-occupancy regime.
+__global__ __launch_bounds__(64) __attribute__((amdgpu_waves_per_eu(1, 1)))
+void k_bls_sigdec_lat(const uint8_t *__restrict__ sigs, uint64_t n,
+                      BlsWork w) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  g2a sig;
+  if (g2_decompress(sig, sigs + 96 * i) != 0) {
+    atomicOr(w.fail, 1);
+    sig.inf = 1;
+  }
+  g2j out;
+  if (sig.inf) {
+    fp2_zero(out.x);
+    fp2_zero(out.y);
+    fp2_zero(out.z);
+  } else {
+    out.x = sig.x;
+    out.y = sig.y;
+    fp2_one(out.z);
+  }
+  w.sig_aff[i] = out;
+}
+
+__global__ __launch_bounds__(64) __attribute__((amdgpu_waves_per_eu(2, 2)))
+void k_bls_prep_mults_lat(const uint8_t *__restrict__ pks,
+                          const uint32_t *__restrict__ offs,
+                          const uint64_t *__restrict__ rands, uint64_t n,
+                          BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 2 * n) return;
+  if (lane < n) {
+    uint64_t i = lane;
+    g2j sj = w.sig_aff[i];
+    if (g2j_is_inf(sj)) {
+      fp2_zero(w.rsig[i].x);
+      fp2_zero(w.rsig[i].y);
+      fp2_zero(w.rsig[i].z);
+      return;
+    }
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j rs;
+    g2_mul_u64_aff(rs, sig, rands[i]);
+    w.rsig[i] = rs;
+    return;
+  }
+  uint64_t i = lane - n;
+  uint32_t k0 = offs[i], k1 = offs[i + 1];
+  if (k1 <= k0) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  g1j rp;
+  if (k1 - k0 == 1) {
+    g1a pk;
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k0) != 0) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    if (pk.inf) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    g1_mul_u64_aff(rp, pk, rands[i]);
+  } else {
+    g1j apk = w.apk[i];
+    if (g1j_is_inf(apk)) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    uint8_t rbe[8];
+#pragma unroll
+    for (int b = 0; b < 8; b++)
+      rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
+    g1j_mul_be_j(rp, apk, rbe, 8);
+  }
+  w.p_scaled[i] = rp;
+  g2j sj = w.sig_aff[i];
+  if (!g2j_is_inf(sj)) {
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j xsig_j;
+    g2_mul_u64_aff(xsig_j, sig, BLS_X_ABS);
+    g2a ps;
+    psi_g2(ps, sig);
+    bool ok;
+    if (g2j_is_inf(xsig_j)) {
+      ok = false;
+    } else {
+      fp2 z2, z3, lx, ly, ny;
+      fp2_sqr(z2, xsig_j.z);
+      fp2_mul(z3, z2, xsig_j.z);
+      fp2_mul(lx, ps.x, z2);
+      fp2_neg(ny, xsig_j.y);
+      fp2_mul(ly, ps.y, z3);
+      ok = fp2_eq(lx, xsig_j.x) && fp2_eq(ly, ny);
+    }
+    if (!ok) atomicOr(w.fail, 1);
+  }
+}
+
+__global__ __launch_bounds__(64) __attribute__((amdgpu_waves_per_eu(2, 2)))
+void k_bls_h2c_map_lat(uint64_t n, BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 2 * n) return;
+  int role = lane < n ? 0 : 1;
+  uint64_t i = role == 0 ? lane : lane - n;
+  const uint8_t *uni = w.uni + 256 * i + 128 * role;
+  fp2 u;
+  h2f_from_be64(u.c0, uni);
+  h2f_from_be64(u.c1, uni + 64);
+  g2a q;
+  sswu_g2(q, u);
+  g2j pt, cleared;
+  iso_map_g2_j(pt, q);
+  clear_cofactor_g2j(cleared, pt);
+  w.h2c_pts[lane] = cleared;
+}
+
 int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                const void *pks_dev, const void *offs_dev,
                const void *rands_dev, uint64_t n, int32_t *out) {
@@ -690,11 +819,12 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
                        (const uint8_t *)pks_dev, (const uint32_t *)offs_dev,
                        (const uint64_t *)rands_dev, n, w);
   } else {
-    // latency regime: decompress pass + two wave-uniform mult classes
-    hipLaunchKernelGGL(k_bls_sigdec, dim3(blocks), dim3(64), 0, ctx->stream,
-                       (const uint8_t *)sigs_dev, n, w);
+    // latency regime: decompress pass + two wave-uniform mult classes,
+    // full-register-budget twins
+    hipLaunchKernelGGL(k_bls_sigdec_lat, dim3(blocks), dim3(64), 0,
+                       ctx->stream, (const uint8_t *)sigs_dev, n, w);
     uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
-    hipLaunchKernelGGL(k_bls_prep_mults, dim3(blocks2), dim3(64), 0,
+    hipLaunchKernelGGL(k_bls_prep_mults_lat, dim3(blocks2), dim3(64), 0,
                        ctx->stream, (const uint8_t *)pks_dev,
                        (const uint32_t *)offs_dev,
                        (const uint64_t *)rands_dev, n, w);
@@ -708,8 +838,12 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
     uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
     hipLaunchKernelGGL(k_bls_h2c_expand, dim3(blocks), dim3(64), 0,
                        ctx->stream2, (const uint8_t *)msgs_dev, n, w);
-    hipLaunchKernelGGL(k_bls_h2c_map, dim3(blocks2), dim3(64), 0,
-                       ctx->stream2, n, w);
+    if (n <= (1ull << 18))
+      hipLaunchKernelGGL(k_bls_h2c_map_lat, dim3(blocks2), dim3(64), 0,
+                         ctx->stream2, n, w);
+    else
+      hipLaunchKernelGGL(k_bls_h2c_map, dim3(blocks2), dim3(64), 0,
+                         ctx->stream2, n, w);
     hipLaunchKernelGGL(k_bls_h2c_fin, dim3(blocks), dim3(64), 0,
                        ctx->stream2, n, w);
   } else {
