@@ -834,7 +834,8 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
   // h2c is independent of prepare: run it on the second stream so the two
   // ~1-wave/SIMD kernels co-reside (both fit at 2 waves/SIMD by VGPR count)
   m3x::time_begin_s(ctx, M3X_K_BLS_H2C, ctx->stream2);
-  if (n > 2048) {
+  if (n > 64) { // split h2c pays at small n too (per-lane chain is ONE
+                // point + clear, vs two points + clear in the fused form)
     uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
     hipLaunchKernelGGL(k_bls_h2c_expand, dim3(blocks), dim3(64), 0,
                        ctx->stream2, (const uint8_t *)msgs_dev, n, w);
