@@ -738,6 +738,96 @@ void k_bls_prep_mults_lat(const uint8_t *__restrict__ pks,
   }
 }
 
+// THREE-class latency-regime mult pass (3n lanes): A [r]sigma,
+// B [|x|]sigma + psi subgroup check, C [r]apk on G1. Same total work as
+// the 2-class form but the critical lane shrinks from (G1 mult + x-mult)
+// to max(one chain) and wave count rises 3n/64 — pure latency win for
+// small/medium batches.
+__global__ __launch_bounds__(64) __attribute__((amdgpu_waves_per_eu(2, 2)))
+void k_bls_prep_mults3(const uint8_t *__restrict__ pks,
+                       const uint32_t *__restrict__ offs,
+                       const uint64_t *__restrict__ rands, uint64_t n,
+                       BlsWork w) {
+  uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (lane >= 3 * n) return;
+  int cls = lane < n ? 0 : (lane < 2 * n ? 1 : 2);
+  uint64_t i = lane - (uint64_t)cls * n;
+  if (cls == 0) { // A: rsig[i] = [r_i] sigma
+    g2j sj = w.sig_aff[i];
+    if (g2j_is_inf(sj)) {
+      fp2_zero(w.rsig[i].x);
+      fp2_zero(w.rsig[i].y);
+      fp2_zero(w.rsig[i].z);
+      return;
+    }
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j rs;
+    g2_mul_u64_aff(rs, sig, rands[i]);
+    w.rsig[i] = rs;
+    return;
+  }
+  if (cls == 1) { // B: deferred psi subgroup check
+    g2j sj = w.sig_aff[i];
+    if (g2j_is_inf(sj)) return; // infinity is a valid element
+    g2a sig;
+    sig.x = sj.x;
+    sig.y = sj.y;
+    sig.inf = 0;
+    g2j xsig_j;
+    g2_mul_u64_aff(xsig_j, sig, BLS_X_ABS);
+    g2a ps;
+    psi_g2(ps, sig);
+    bool ok;
+    if (g2j_is_inf(xsig_j)) {
+      ok = false;
+    } else {
+      fp2 z2, z3, lx, ly, ny;
+      fp2_sqr(z2, xsig_j.z);
+      fp2_mul(z3, z2, xsig_j.z);
+      fp2_mul(lx, ps.x, z2);
+      fp2_neg(ny, xsig_j.y);
+      fp2_mul(ly, ps.y, z3);
+      ok = fp2_eq(lx, xsig_j.x) && fp2_eq(ly, ny);
+    }
+    if (!ok) atomicOr(w.fail, 1);
+    return;
+  }
+  // C: p_scaled[i] = [r_i] apk
+  uint32_t k0 = offs[i], k1 = offs[i + 1];
+  if (k1 <= k0) {
+    atomicOr(w.fail, 1);
+    return;
+  }
+  g1j rp;
+  if (k1 - k0 == 1) {
+    g1a pk;
+    if (g1_from_uncomp_trusted(pk, pks + 96 * (uint64_t)k0) != 0) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    if (pk.inf) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    g1_mul_u64_aff(rp, pk, rands[i]);
+  } else {
+    g1j apk = w.apk[i];
+    if (g1j_is_inf(apk)) {
+      atomicOr(w.fail, 1);
+      return;
+    }
+    uint8_t rbe[8];
+#pragma unroll
+    for (int b = 0; b < 8; b++)
+      rbe[b] = (uint8_t)(rands[i] >> (56 - 8 * b));
+    g1j_mul_be_j(rp, apk, rbe, 8);
+  }
+  w.p_scaled[i] = rp;
+}
+
 __global__ __launch_bounds__(64) __attribute__((amdgpu_waves_per_eu(2, 2)))
 void k_bls_h2c_map_lat(uint64_t n, BlsWork w) {
   uint64_t lane = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -823,8 +913,8 @@ int run_verify(m3x_ctx *ctx, const void *msgs_dev, const void *sigs_dev,
     // full-register-budget twins
     hipLaunchKernelGGL(k_bls_sigdec_lat, dim3(blocks), dim3(64), 0,
                        ctx->stream, (const uint8_t *)sigs_dev, n, w);
-    uint32_t blocks2 = (uint32_t)((2 * n + 63) / 64);
-    hipLaunchKernelGGL(k_bls_prep_mults_lat, dim3(blocks2), dim3(64), 0,
+    uint32_t blocks3 = (uint32_t)((3 * n + 63) / 64);
+    hipLaunchKernelGGL(k_bls_prep_mults3, dim3(blocks3), dim3(64), 0,
                        ctx->stream, (const uint8_t *)pks_dev,
                        (const uint32_t *)offs_dev,
                        (const uint64_t *)rands_dev, n, w);
