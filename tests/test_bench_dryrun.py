@@ -4,7 +4,7 @@ Launches bench.py under torch.distributed.run with the gloo backend and
 --dry-run: compute is stubbed, but the collective code the driver's
 8-GPU SCALE bench will execute — verdict all_reduce(MIN), subtree-root
 all_gather + rank-0 cap finishing, barriers, max-over-ranks timing,
-rank-0 JSON emit — runs verbatim at world sizes 2 and 8.
+rank-0 JSON emit — runs verbatim at world sizes 2, 4 and 8.
 """
 import json
 import subprocess
@@ -16,7 +16,7 @@ import pytest
 REPO = Path(__file__).resolve().parent.parent
 
 
-@pytest.mark.parametrize("world", [2, 8])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_bench_dry_run_world(world):
     port = 29580 + world
     cmd = [
