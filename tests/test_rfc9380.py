@@ -213,3 +213,27 @@ def test_gpu_non_subgroup_signature_rejected(ctx):
     ]
     rands = [int(r) for r in case["rands"]]
     assert bls.verify_signature_sets(sets, ctx=ctx, _rands=rands) is True
+
+
+def test_oracle_expand_xmd_fuzz_vs_python(oracle):
+    """General-DST expander vs the independent Python implementation
+    (tests/golden/gen_bls_fixtures.py, itself pinned by the literal RFC
+    vectors above) across random messages, DSTs and output lengths."""
+    import random
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).parent / "golden"))
+    import gen_bls_fixtures as ref
+
+    rng = random.Random(0xC0FFEE)
+    for _ in range(40):
+        msg = bytes(rng.randrange(256) for _ in range(rng.randrange(0, 200)))
+        dst = bytes(rng.randrange(1, 127) for _ in range(rng.randrange(1, 255)))
+        length = rng.choice([1, 16, 32, 33, 64, 96, 128, 255, 256])
+        want = ref.expand_message_xmd(msg, dst, length)
+        out = ctypes.create_string_buffer(length)
+        oracle.m3x_oracle_expand_xmd(
+            msg, len(msg), dst, len(dst), length, out
+        )
+        assert out.raw == bytes(want), (msg.hex()[:16], dst.hex()[:16], length)
