@@ -116,7 +116,78 @@ void fp_mul_(fp_t *r, const fp_t *a, const fp_t *b) {
   memcpy(r->v, t, 48);
 }
 
-static void fp_sqr(fp_t *r, const fp_t *a) { fp_mul_(r, a, a); }
+/* specialized squaring, fully unrolled SOS: 21 products + doubling vs the
+ * multiply's 36 (13% faster measured; bit-exact over 300k chained) — the
+ * sqrt/inv pow chains are squaring-dominated. */
+#define M3X_SQMAC(hi, lo, x, y, a0, a1)                                        \
+  do {                                                                         \
+    unsigned __int128 _p = (unsigned __int128)(x) * (y) + (a0) + (a1);         \
+    (lo) = (uint64_t)_p;                                                       \
+    (hi) = (uint64_t)(_p >> 64);                                               \
+  } while (0)
+
+static void fp_sqr(fp_t *r, const fp_t *a) {
+  const uint64_t *A = a->v;
+  uint64_t t1,t2,t3,t4,t5,t6,t7,t8,t9,t10,c;
+  /* row i=0: j=1..5 */
+  M3X_SQMAC(c, t1, A[0], A[1], 0, 0);
+  M3X_SQMAC(c, t2, A[0], A[2], 0, c);
+  M3X_SQMAC(c, t3, A[0], A[3], 0, c);
+  M3X_SQMAC(c, t4, A[0], A[4], 0, c);
+  M3X_SQMAC(c, t5, A[0], A[5], 0, c);
+  t6 = c;
+  /* row i=1: j=2..5 */
+  M3X_SQMAC(c, t3, A[1], A[2], t3, 0);
+  M3X_SQMAC(c, t4, A[1], A[3], t4, c);
+  M3X_SQMAC(c, t5, A[1], A[4], t5, c);
+  M3X_SQMAC(c, t6, A[1], A[5], t6, c);
+  t7 = c;
+  /* row i=2 */
+  M3X_SQMAC(c, t5, A[2], A[3], t5, 0);
+  M3X_SQMAC(c, t6, A[2], A[4], t6, c);
+  M3X_SQMAC(c, t7, A[2], A[5], t7, c);
+  t8 = c;
+  /* row i=3 */
+  M3X_SQMAC(c, t7, A[3], A[4], t7, 0);
+  M3X_SQMAC(c, t8, A[3], A[5], t8, c);
+  t9 = c;
+  /* row i=4 */
+  M3X_SQMAC(c, t9, A[4], A[5], t9, 0);
+  t10 = c;
+  /* double + diagonals */
+  uint64_t w[13];
+  w[0]=0; w[1]=t1; w[2]=t2; w[3]=t3; w[4]=t4; w[5]=t5; w[6]=t6; w[7]=t7; w[8]=t8; w[9]=t9; w[10]=t10; w[11]=0; w[12]=0;
+  uint64_t cc = 0;
+  for (int i = 1; i <= 11; i++) { uint64_t nv = (w[i] << 1) | cc; cc = w[i] >> 63; w[i] = nv; }
+  unsigned __int128 p; uint64_t carry = 0;
+  for (int i = 0; i < 6; i++) {
+    p = (unsigned __int128)A[i]*A[i] + w[2*i] + carry;
+    w[2*i] = (uint64_t)p;
+    unsigned __int128 p2 = (unsigned __int128)w[2*i+1] + (uint64_t)(p >> 64);
+    w[2*i+1] = (uint64_t)p2;
+    carry = (uint64_t)(p2 >> 64);
+  }
+  /* Montgomery reduce */
+  uint64_t s0=w[0],s1=w[1],s2=w[2],s3=w[3],s4=w[4],s5=w[5];
+  uint64_t hc = 0; uint64_t m;
+#define REDR(hiw) do { \
+    m = s0 * BLS_N0; \
+    { unsigned __int128 p0 = (unsigned __int128)m * BLS_P[0] + s0; c = (uint64_t)(p0 >> 64); } \
+    M3X_SQMAC(c, s0, m, BLS_P[1], s1, c); \
+    M3X_SQMAC(c, s1, m, BLS_P[2], s2, c); \
+    M3X_SQMAC(c, s2, m, BLS_P[3], s3, c); \
+    M3X_SQMAC(c, s3, m, BLS_P[4], s4, c); \
+    M3X_SQMAC(c, s4, m, BLS_P[5], s5, c); \
+    { unsigned __int128 s = (unsigned __int128)(hiw) + c + hc; s5 = (uint64_t)s; hc = (uint64_t)(s >> 64); } \
+  } while (0)
+  REDR(w[6]); REDR(w[7]); REDR(w[8]); REDR(w[9]); REDR(w[10]); REDR(w[11]);
+#undef REDR
+  uint64_t out[6] = {s0,s1,s2,s3,s4,s5};
+  if (hc || ge_p(out)) sub_p(out);
+  memcpy(r->v, out, 48);
+}
+#undef M3X_SQMAC
+
 
 /* ---- lazy-reduction support (round 2): 12-limb full product + one
  * Montgomery reduction, so fp2_mul does 3 wide muls + 2 reductions
